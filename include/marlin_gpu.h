@@ -1,0 +1,160 @@
+/* marlin_gpu.h — C ABI of the MI355X-native block-matrix multiply engine.
+ *
+ * This is the drop-in boundary replacing the reference's (PasaLab/marlin)
+ * native hand-off point: Breeze `*` -> netlib-java JNI `dgemm` invoked at
+ *   SubMatrix.scala:87-105  (per-tile dense multiply)
+ *   SubMatrix.scala:41-50   (per-tile partial-sum add, reduceByKey combiner)
+ * and the whole-multiply operator surface
+ *   BlockMatrix.scala:149-220  (BlockMatrix.multiply)
+ *   DenseVecMatrix.scala:109-141, 196-231 (split-mode multiply + dispatch).
+ *
+ * A JVM host (Marlin's Scala DenseVecMatrix/BlockMatrix) binds these entry
+ * points over JNI (see INTEGRATION.md for the binding stub); the in-container
+ * hosts are src/host/marlinx.cpp (C++ CLI) and marlin_amd/engine.py (ctypes).
+ *
+ * Conventions (match the reference):
+ *   - All matrices are COLUMN-MAJOR fp64/fp32 (Breeze's layout,
+ *     Matrices.scala:34-48), caller-owned host buffers, fully materialised.
+ *   - All functions return int: 0 = ok, < 0 = error (mx_strerror).
+ *   - Dimension checks mirror `require(numCols == other.numRows)`
+ *     (BlockMatrix.scala:150-151) -> MX_EDIM.
+ *   - An mx_ctx is externally synchronised (one call at a time — matches
+ *     Spark's one-action-at-a-time driver); internally multi-stream.
+ */
+#ifndef MARLIN_GPU_H
+#define MARLIN_GPU_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- error codes ------------------------------------------------------- */
+#define MX_OK            0
+#define MX_EDIM         -1   /* dimension mismatch (require(...) analogue)  */
+#define MX_EHIP         -2   /* HIP runtime failure                          */
+#define MX_ENOMEM       -3   /* device allocation failure                    */
+#define MX_EINVAL       -4   /* bad argument (null ptr, size<=0, bad grid)   */
+#define MX_ENOCOMM      -5   /* distributed entry without mx_comm_init       */
+#define MX_ERCCL        -6   /* RCCL failure                                 */
+#define MX_ENODEV       -7   /* no MI355X device visible                     */
+
+const char* mx_strerror(int code);
+
+/* ---- context ----------------------------------------------------------- */
+typedef struct mx_ctx mx_ctx;
+
+/* Create a context bound to one GPU (one process per GPU).
+ * device < 0 means "current HIP device". */
+int mx_init(mx_ctx** out, int device);
+int mx_shutdown(mx_ctx* ctx);
+
+/* ---- distributed setup (SUMMA over RCCL/xGMI) --------------------------- */
+/* 128-byte opaque RCCL unique id, exchanged out-of-band by the launcher
+ * (bench.py uses torch.distributed/gloo as the control plane). */
+#define MX_UNIQUE_ID_BYTES 128
+int mx_comm_id(char unique_id[MX_UNIQUE_ID_BYTES]);  /* rank 0 calls this */
+
+/* Collective: every rank of the job calls with the same unique_id.
+ * Builds the world communicator plus the pr x pc grid row/col
+ * sub-communicators (grids: 8=4x2, 4=2x2, 2=2x1, 1=1x1). */
+int mx_comm_init(mx_ctx* ctx, int rank, int nranks,
+                 const char unique_id[MX_UNIQUE_ID_BYTES]);
+
+/* Grid geometry of this rank after mx_comm_init (pr,pc,row,col). */
+int mx_grid(mx_ctx* ctx, int* pr, int* pc, int* prow, int* pcol);
+
+/* ---- whole-multiply entries (the BlockMatrix.multiply replacement) ------ */
+/* Single-GPU: C = A * B, host col-major buffers; the engine does
+ * pad/H2D/tiled MFMA GEMM/D2H internally. */
+int mx_dgemm(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
+             const double* A, const double* B, double* C);
+int mx_sgemm(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
+             const float* A, const float* B, float* C);
+
+/* fp32 multiply with fused epilogue (BASELINE config 5):
+ * C_out = op(A*B) [+ add_c], op = transpose if transpose_c != 0.
+ * If transpose_c: C is n x m col-major, add_c (optional, may be NULL)
+ * n x m; else C is m x n. Epilogue runs on-device, fused into the
+ * result store — replaces BlockMatrix transpose (BlockMatrix.scala:514-523)
+ * + elementwise add (BlockMatrix.scala:344-452) composed after multiply. */
+int mx_sgemm_epilogue(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
+                      const float* A, const float* B, float* C,
+                      int transpose_c, const float* add_c);
+
+/* Distributed SUMMA (replaces the Spark shuffle route,
+ * BlockMatrix.scala:161-186): every rank passes ITS OWN shards.
+ * Layout (block distribution on the pr x pc grid):
+ *   A_local: ceil-rows slab  [my_m x k]   (rows  owner: prow)
+ *   B_local: ceil-cols slab  [k x my_n]   (cols  owner: pcol)
+ *   C_local: [my_m x my_n]
+ * my_m = slab_len(m, pr, prow), my_n = slab_len(n, pc, pcol) — ceil split,
+ * last slab ragged (DenseVecMatrix.scala:1262-1265 semantics).
+ * Panels of A (within grid rows) and B (within grid cols) are broadcast
+ * per k-step over RCCL/xGMI, double-buffered against the MFMA stream. */
+int mx_dgemm_summa(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
+                   const double* A_local, const double* B_local,
+                   double* C_local);
+int mx_sgemm_summa(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
+                   const float* A_local, const float* B_local,
+                   float* C_local);
+
+/* Ceil slab split helper (exact reference blocking semantics). */
+int64_t mx_slab_len(int64_t total, int parts, int idx);
+int64_t mx_slab_off(int64_t total, int parts, int idx);
+
+/* ---- per-tile entries (the SubMatrix.multiply/add replacement) ----------
+ * For a host that still does its own blocking (Marlin's Scala layer):
+ * device-resident accumulate-GEMM per tile. beta_one=0: C=A*B;
+ * beta_one=1: C+=A*B (the SubMatrix.add combiner folded in). */
+int mx_tile_dgemm_acc(mx_ctx* ctx, int64_t tm, int64_t tk, int64_t tn,
+                      const double* hA_tile, const double* hB_tile,
+                      double* hC_tile, int beta_one);
+
+/* ---- device-resident entries (buffers already in HBM; used by bench) ---- */
+typedef struct mx_dbuf mx_dbuf;          /* opaque device buffer            */
+int mx_alloc(mx_ctx* ctx, int64_t bytes, mx_dbuf** out);
+int mx_free(mx_ctx* ctx, mx_dbuf* buf);
+int mx_upload(mx_ctx* ctx, mx_dbuf* dst, const void* src, int64_t bytes);
+int mx_download(mx_ctx* ctx, void* dst, const mx_dbuf* src, int64_t bytes);
+/* Fill a device fp64 buffer with the deterministic xorshift64* U[0,1)
+ * stream (the randomDenVecMatrix stand-in, MTUtils.scala:63-73 semantics:
+ * seeded, per-element reproducible). */
+int mx_fill_random(mx_ctx* ctx, mx_dbuf* buf, int64_t n_elems,
+                   uint64_t seed, int is_fp32);
+/* Device-resident GEMM on padded-pitch device buffers created by the
+ * helpers above. lda/ldb/ldc are element pitches. */
+int mx_dgemm_device(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
+                    const mx_dbuf* dA, int64_t lda,
+                    const mx_dbuf* dB, int64_t ldb,
+                    mx_dbuf* dC, int64_t ldc);
+int mx_sgemm_device(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
+                    const mx_dbuf* dA, int64_t lda,
+                    const mx_dbuf* dB, int64_t ldb,
+                    mx_dbuf* dC, int64_t ldc);
+/* SUMMA on device-resident local shards (bench hot loop: inputs already
+ * in HBM when the timed region starts). */
+int mx_dgemm_summa_device(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
+                          const mx_dbuf* dA_local, const mx_dbuf* dB_local,
+                          mx_dbuf* dC_local);
+
+/* ---- timing / stats (MTUtils.evaluate + RMMcompare.scala:47-51 analog) -- */
+typedef struct {
+  double h2d_ms;          /* host->device copies of the last call           */
+  double d2h_ms;          /* device->host copies                            */
+  double pack_ms;         /* padding/pack kernels                           */
+  double gemm_ms;         /* sum of dgemm kernel time (hipEvent, gemm strm) */
+  double comm_ms;         /* RCCL panel broadcast time (comm stream)        */
+  double total_ms;        /* wall of the whole entry                        */
+  int64_t gemm_launches;  /* number of GEMM kernel launches                 */
+  double flops;           /* algorithmic flops of the last call (2mkn)      */
+  double bytes_moved;     /* algorithmic HBM bytes (inputs+outputs, padded) */
+} mx_stats_t;
+int mx_stats(mx_ctx* ctx, mx_stats_t* out);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* MARLIN_GPU_H */

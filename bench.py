@@ -1,0 +1,206 @@
+#!/usr/bin/env python3
+# bench.py — headline benchmark of the MI355X block-matrix multiply engine.
+#
+# Metric (BASELINE.json): dense C=A*B TFLOP/s (fp64) at N=20000.
+# A "step" = one full C = A x B over device-resident inputs (inputs are in
+# HBM before the timed region starts; H2D is NOT in the timed region and
+# the PCIe-inclusive rate is reported in DESIGN.md).
+#
+#   python bench.py --gpus N --steps K --warmup W
+#
+# N > 1 is launched by the driver via torch.distributed.run (one rank per
+# GPU); ranks use gloo ONLY as the control plane (RCCL unique-id exchange,
+# barriers) — the data path is the engine's own RCCL-over-xGMI SUMMA.
+# Rank 0 prints ONE JSON line.
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, HERE)
+
+from marlin_amd import Engine              # noqa: E402
+from marlin_amd import engine as eng_mod   # noqa: E402
+
+FP64_PEAK_TF = 78.6   # gfx950: 256 CU x 4 SIMD x 32 FLOP/clk x 2.4 GHz
+FP32_PEAK_TF = 157.3
+
+
+def roundup(x, a):
+    return (x + a - 1) // a * a
+
+
+def cpu_baseline_leg(args):
+    """Oracle restatement (the reference's blocked algorithm, OpenBLAS
+    per-tile dgemm) timed on the host cores — BASELINE.md plan. Bounded
+    sample: nb^3 multiply (~seconds of CPU), scaled to TFLOP/s."""
+    from oracle import gen_matrix, blocked_multiply, split_method
+    cores = os.cpu_count() or 1
+    nb = int(args.cpu_sample)
+    a = gen_matrix(nb, nb, seed=0xA11CE)
+    b = gen_matrix(nb, nb, seed=0xB0B)
+    mkn = split_method(nb, nb, nb, cores)
+    t0 = time.perf_counter()
+    blocked_multiply(a, b, mkn)
+    dt = time.perf_counter() - t0
+    tf = 2.0 * nb ** 3 / dt / 1e12
+    return {
+        "value": round(tf, 4), "unit": "TFLOP/s", "cores": cores,
+        "kind": "port",
+        "sample": f"{nb}^3 fp64 blocked multiply (CARMA split "
+                  f"{mkn}, OpenBLAS tiles, {dt:.2f}s)",
+    }
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--n", type=int, default=20000,
+                   help="square problem size (m=k=n)")
+    p.add_argument("--m", type=int, default=0)
+    p.add_argument("--k", type=int, default=0)
+    p.add_argument("--nn", type=int, default=0)
+    p.add_argument("--dtype", choices=["f64", "f32"], default="f64")
+    p.add_argument("--cpu-sample", type=int, default=2048)
+    p.add_argument("--no-cpu-baseline", action="store_true")
+    args = p.parse_args()
+
+    m = args.m or args.n
+    k = args.k or args.n
+    n = args.nn or args.n
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        tdist.init_process_group(backend="gloo", rank=rank, world_size=world)
+        dist = tdist
+
+    fp32 = args.dtype == "f32"
+    elem = 4 if fp32 else 8
+    eng = Engine(local_rank)
+
+    # --- distributed setup + device-resident inputs ----------------------
+    if world > 1:
+        import torch
+        if rank == 0:
+            uid = Engine.comm_id()
+            t = torch.tensor(list(uid), dtype=torch.uint8)
+        else:
+            t = torch.zeros(eng_mod.UNIQUE_ID_BYTES, dtype=torch.uint8)
+        dist.broadcast(t, src=0)
+        eng.comm_init(rank, world, bytes(t.tolist()))
+        pr, pc, prow, pcol = eng.grid()
+    else:
+        pr = pc = 1
+        prow = pcol = 0
+
+    mi = eng_mod.slab_len(m, pr, prow)
+    nj = eng_mod.slab_len(n, pc, pcol)
+    kaj = eng_mod.slab_len(k, pc, pcol)
+    kbi = eng_mod.slab_len(k, pr, prow)
+    mip, njp = roundup(mi, 128), roundup(nj, 128)
+    kbi_p = roundup(kbi, 16)
+
+    if world > 1:
+        dA = eng.alloc(mip * kaj * elem)
+        dB = eng.alloc(kbi_p * nj * elem)
+        dC = eng.alloc(mip * njp * elem)
+        eng.fill_random(dA, mip * kaj, 0xA11CE + rank, fp32)
+        eng.fill_random(dB, kbi_p * nj, 0xB0B + rank, fp32)
+
+        def step():
+            eng.dgemm_summa_device(m, k, n, dA, dB, dC)
+    else:
+        mp, kp, np_ = roundup(m, 128), roundup(k, 16), roundup(n, 128)
+        dA = eng.alloc(mp * kp * elem)
+        dB = eng.alloc(kp * np_ * elem)
+        dC = eng.alloc(mp * np_ * elem)
+        eng.fill_random(dA, mp * kp, 0xA11CE, fp32)
+        eng.fill_random(dB, kp * np_, 0xB0B, fp32)
+        gem = eng.sgemm_device if fp32 else eng.dgemm_device
+
+        def step():
+            gem(mp, kp, np_, dA, mp, dB, kp, dC, mp)
+
+    # --- warmup / timed region ------------------------------------------
+    for _ in range(args.warmup):
+        step()
+    if dist:
+        dist.barrier()
+    t0 = time.perf_counter()
+    gemm_ms_acc = 0.0
+    launches = 0
+    for _ in range(args.steps):
+        step()
+        st = eng.stats()
+        gemm_ms_acc += st["gemm_ms"]
+        launches += st["gemm_launches"]
+    # engine entries synchronize internally; no extra device sync needed
+    dt = time.perf_counter() - t0
+    if dist:
+        import torch
+        tmax = torch.tensor([dt])
+        dist.all_reduce(tmax, op=dist.ReduceOp.MAX)
+        dt = float(tmax[0])
+
+    if rank == 0:
+        flops = 2.0 * m * k * n * args.steps
+        tf = flops / dt / 1e12
+        ms_per_step = dt * 1000.0 / args.steps
+        peak = FP32_PEAK_TF if fp32 else FP64_PEAK_TF
+        # dominant-kernel roofline: HIP-event time of the MFMA GEMM kernel
+        # launches (stats from the engine's gemm-stream events)
+        if launches and gemm_ms_acc > 0:
+            ach = (2.0 * m * k * n * args.steps) / (gemm_ms_acc / 1e3) / 1e12
+        else:
+            ach = None
+        roofline = {
+            "bound": "mfma",
+            "achieved": round(ach, 3) if ach else None,
+            "peak": peak * world,
+            "unit": "TFLOP/s",
+            "frac": round(ach / (peak * world), 4) if ach else None,
+            "traffic": None,  # rocprofv3 PMC summaries: profiles/
+        }
+        out = {
+            "metric": "dense C=AxB TFLOP/s (fp64)" if not fp32 else
+                      "dense C=AxB TFLOP/s (fp32)",
+            "value": round(tf, 3),
+            "unit": "TFLOP/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "workload": f"dgemm_{m}x{k}x{n}_{args.dtype}",
+                "m": m, "k": k, "n": n,
+                "parallelism": f"summa_grid_{pr}x{pc}" if world > 1 else
+                               "single_gpu",
+            },
+            "roofline": roofline,
+        }
+        if not args.no_cpu_baseline and world == 1:
+            out["cpu_baseline"] = cpu_baseline_leg(args)
+        print(json.dumps(out), flush=True)
+    if dist:
+        dist.destroy_process_group()
+    eng.close()
+
+
+if __name__ == "__main__":
+    main()

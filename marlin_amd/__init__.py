@@ -1,0 +1,5 @@
+# marlin_amd — MI355X-native block-matrix multiply engine.
+# Product path: hand-written HIP/CDNA4 MFMA kernels + RCCL over xGMI
+# behind the C ABI of include/marlin_gpu.h. No CPU fallback anywhere.
+from .engine import Engine, EngineError, EngineUnavailable  # noqa: F401
+from .api import DenseVecMatrix, BlockMatrix, BlockID, split_method  # noqa: F401

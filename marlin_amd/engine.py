@@ -1,0 +1,288 @@
+# engine.py — ctypes binding to libmarlin_gpu.so (the C ABI of
+# include/marlin_gpu.h). This is the product compute path: it NEVER falls
+# back to CPU. If the extension or a GPU is missing, every compute entry
+# raises EngineUnavailable loudly.
+import ctypes
+import os
+
+import numpy as np
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_HERE, "libmarlin_gpu.so")
+
+UNIQUE_ID_BYTES = 128
+
+_ERRNAMES = {
+    0: "MX_OK", -1: "MX_EDIM", -2: "MX_EHIP", -3: "MX_ENOMEM",
+    -4: "MX_EINVAL", -5: "MX_ENOCOMM", -6: "MX_ERCCL", -7: "MX_ENODEV",
+}
+
+
+class EngineUnavailable(RuntimeError):
+    pass
+
+
+class EngineError(RuntimeError):
+    def __init__(self, code, what=""):
+        super().__init__(f"marlin_gpu error {_ERRNAMES.get(code, code)} {what}")
+        self.code = code
+
+
+class MxStats(ctypes.Structure):
+    _fields_ = [
+        ("h2d_ms", ctypes.c_double), ("d2h_ms", ctypes.c_double),
+        ("pack_ms", ctypes.c_double), ("gemm_ms", ctypes.c_double),
+        ("comm_ms", ctypes.c_double), ("total_ms", ctypes.c_double),
+        ("gemm_launches", ctypes.c_int64), ("flops", ctypes.c_double),
+        ("bytes_moved", ctypes.c_double),
+    ]
+
+
+def _load():
+    if not os.path.exists(_SO):
+        raise EngineUnavailable(
+            f"HIP extension not built: {_SO} missing. Run "
+            f"python -c 'import __graft_entry__; __graft_entry__.build()'")
+    try:
+        lib = ctypes.CDLL(_SO)
+    except OSError as e:
+        raise EngineUnavailable(f"cannot load {_SO}: {e}")
+    P = ctypes.POINTER
+    i64, u64, dbl, flt = (ctypes.c_int64, ctypes.c_uint64, ctypes.c_double,
+                          ctypes.c_float)
+    vp = ctypes.c_void_p
+    sigs = {
+        "mx_strerror": (ctypes.c_char_p, [ctypes.c_int]),
+        "mx_init": (ctypes.c_int, [P(vp), ctypes.c_int]),
+        "mx_shutdown": (ctypes.c_int, [vp]),
+        "mx_comm_id": (ctypes.c_int, [ctypes.c_char_p]),
+        "mx_comm_init": (ctypes.c_int, [vp, ctypes.c_int, ctypes.c_int,
+                                        ctypes.c_char_p]),
+        "mx_grid": (ctypes.c_int, [vp] + [P(ctypes.c_int)] * 4),
+        "mx_dgemm": (ctypes.c_int, [vp, i64, i64, i64, P(dbl), P(dbl), P(dbl)]),
+        "mx_sgemm": (ctypes.c_int, [vp, i64, i64, i64, P(flt), P(flt), P(flt)]),
+        "mx_sgemm_epilogue": (ctypes.c_int, [vp, i64, i64, i64, P(flt), P(flt),
+                                             P(flt), ctypes.c_int, P(flt)]),
+        "mx_tile_dgemm_acc": (ctypes.c_int, [vp, i64, i64, i64, P(dbl), P(dbl),
+                                             P(dbl), ctypes.c_int]),
+        "mx_dgemm_summa": (ctypes.c_int, [vp, i64, i64, i64, P(dbl), P(dbl),
+                                          P(dbl)]),
+        "mx_sgemm_summa": (ctypes.c_int, [vp, i64, i64, i64, P(flt), P(flt),
+                                          P(flt)]),
+        "mx_slab_len": (i64, [i64, ctypes.c_int, ctypes.c_int]),
+        "mx_slab_off": (i64, [i64, ctypes.c_int, ctypes.c_int]),
+        "mx_plan_panels": (ctypes.c_int, [i64, ctypes.c_int, ctypes.c_int, i64,
+                                          P(i64), P(i64), P(ctypes.c_int),
+                                          P(ctypes.c_int), ctypes.c_int]),
+        "mx_alloc": (ctypes.c_int, [vp, i64, P(vp)]),
+        "mx_free": (ctypes.c_int, [vp, vp]),
+        "mx_upload": (ctypes.c_int, [vp, vp, vp, i64]),
+        "mx_download": (ctypes.c_int, [vp, vp, vp, i64]),
+        "mx_fill_random": (ctypes.c_int, [vp, vp, i64, u64, ctypes.c_int]),
+        "mx_dgemm_device": (ctypes.c_int, [vp, i64, i64, i64, vp, i64, vp, i64,
+                                           vp, i64]),
+        "mx_sgemm_device": (ctypes.c_int, [vp, i64, i64, i64, vp, i64, vp, i64,
+                                           vp, i64]),
+        "mx_dgemm_summa_device": (ctypes.c_int, [vp, i64, i64, i64, vp, vp, vp]),
+        "mx_stats": (ctypes.c_int, [vp, P(MxStats)]),
+    }
+    for name, (res, args) in sigs.items():
+        fn = getattr(lib, name)
+        fn.restype = res
+        fn.argtypes = args
+    return lib
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        _lib = _load()
+    return _lib
+
+
+def _ck(code, what=""):
+    if code != 0:
+        raise EngineError(code, what)
+
+
+# --- pure planning helpers (no GPU needed) ---------------------------------
+def slab_len(total, parts, idx):
+    return lib().mx_slab_len(total, parts, idx)
+
+
+def slab_off(total, parts, idx):
+    return lib().mx_slab_off(total, parts, idx)
+
+
+def plan_panels(K, pr, pc, kb_max=4096):
+    cap = 4096
+    k0 = (ctypes.c_int64 * cap)()
+    k1 = (ctypes.c_int64 * cap)()
+    ra = (ctypes.c_int * cap)()
+    rb = (ctypes.c_int * cap)()
+    npan = lib().mx_plan_panels(K, pr, pc, kb_max, k0, k1, ra, rb, cap)
+    if npan < 0:
+        raise EngineError(-4, f"panel overflow {npan}")
+    return [(k0[i], k1[i], ra[i], rb[i]) for i in range(npan)]
+
+
+def grid_shape(nranks):
+    return {8: (4, 2), 4: (2, 2), 2: (2, 1), 1: (1, 1)}.get(nranks,
+                                                            (nranks, 1))
+
+
+def _fbuf(a, dtype):
+    assert a.dtype == dtype and a.flags.f_contiguous
+    return a.ctypes.data_as(ctypes.POINTER(
+        ctypes.c_double if dtype == np.float64 else ctypes.c_float))
+
+
+class Engine:
+    """One GPU per process. Owns an mx_ctx."""
+
+    def __init__(self, device=-1):
+        self._ctx = ctypes.c_void_p()
+        rc = lib().mx_init(ctypes.byref(self._ctx), device)
+        if rc == -7:
+            raise EngineUnavailable("no MI355X visible (mx_init -> MX_ENODEV)")
+        _ck(rc, "mx_init")
+
+    def close(self):
+        if self._ctx:
+            lib().mx_shutdown(self._ctx)
+            self._ctx = None
+
+    # -- distributed setup --------------------------------------------------
+    @staticmethod
+    def comm_id():
+        buf = ctypes.create_string_buffer(UNIQUE_ID_BYTES)
+        _ck(lib().mx_comm_id(buf), "mx_comm_id")
+        return buf.raw
+
+    def comm_init(self, rank, nranks, uid_bytes):
+        assert len(uid_bytes) == UNIQUE_ID_BYTES
+        _ck(lib().mx_comm_init(self._ctx, rank, nranks, uid_bytes),
+            "mx_comm_init")
+
+    def grid(self):
+        vals = [ctypes.c_int() for _ in range(4)]
+        _ck(lib().mx_grid(self._ctx, *[ctypes.byref(v) for v in vals]))
+        return tuple(v.value for v in vals)  # pr, pc, prow, pcol
+
+    # -- whole multiplies ----------------------------------------------------
+    def dgemm(self, A, B):
+        A = np.asfortranarray(A, dtype=np.float64)
+        B = np.asfortranarray(B, dtype=np.float64)
+        m, k = A.shape
+        k2, n = B.shape
+        if k != k2:
+            raise ValueError(
+                f"Dimension mismatch during matrix-matrix multiplication: "
+                f"{k} vs {k2}")
+        C = np.empty((m, n), dtype=np.float64, order="F")
+        _ck(lib().mx_dgemm(self._ctx, m, k, n, _fbuf(A, np.float64),
+                           _fbuf(B, np.float64), _fbuf(C, np.float64)),
+            "mx_dgemm")
+        return C
+
+    def sgemm(self, A, B):
+        A = np.asfortranarray(A, dtype=np.float32)
+        B = np.asfortranarray(B, dtype=np.float32)
+        m, k = A.shape
+        _, n = B.shape
+        if A.shape[1] != B.shape[0]:
+            raise ValueError("Dimension mismatch during matrix-matrix multiplication")
+        C = np.empty((m, n), dtype=np.float32, order="F")
+        _ck(lib().mx_sgemm(self._ctx, m, k, n, _fbuf(A, np.float32),
+                           _fbuf(B, np.float32), _fbuf(C, np.float32)),
+            "mx_sgemm")
+        return C
+
+    def sgemm_transpose_add(self, A, B, add_c=None):
+        """C = (A*B)^T (+ add_c) with the epilogue fused on-device."""
+        A = np.asfortranarray(A, dtype=np.float32)
+        B = np.asfortranarray(B, dtype=np.float32)
+        m, k = A.shape
+        _, n = B.shape
+        C = np.empty((n, m), dtype=np.float32, order="F")
+        addp = None
+        if add_c is not None:
+            add_c = np.asfortranarray(add_c, dtype=np.float32)
+            assert add_c.shape == (n, m)
+            addp = _fbuf(add_c, np.float32)
+        _ck(lib().mx_sgemm_epilogue(self._ctx, m, k, n, _fbuf(A, np.float32),
+                                    _fbuf(B, np.float32), _fbuf(C, np.float32),
+                                    1, addp), "mx_sgemm_epilogue")
+        return C
+
+    def tile_dgemm_acc(self, A, B, C=None):
+        """C (+)= A*B per tile — the SubMatrix.multiply/add replacement."""
+        A = np.asfortranarray(A, dtype=np.float64)
+        B = np.asfortranarray(B, dtype=np.float64)
+        m, k = A.shape
+        _, n = B.shape
+        beta = 1 if C is not None else 0
+        if C is None:
+            C = np.empty((m, n), dtype=np.float64, order="F")
+        else:
+            C = np.asfortranarray(C, dtype=np.float64)
+        _ck(lib().mx_tile_dgemm_acc(self._ctx, m, k, n, _fbuf(A, np.float64),
+                                    _fbuf(B, np.float64), _fbuf(C, np.float64),
+                                    beta), "mx_tile_dgemm_acc")
+        return C
+
+    def dgemm_summa(self, m, k, n, A_local, B_local):
+        """Distributed multiply on this rank's shards (after comm_init)."""
+        A_local = np.asfortranarray(A_local, dtype=np.float64)
+        B_local = np.asfortranarray(B_local, dtype=np.float64)
+        pr, pc, prow, pcol = self.grid()
+        mi = slab_len(m, pr, prow)
+        nj = slab_len(n, pc, pcol)
+        C = np.empty((mi, nj), dtype=np.float64, order="F")
+        _ck(lib().mx_dgemm_summa(self._ctx, m, k, n, _fbuf(A_local, np.float64),
+                                 _fbuf(B_local, np.float64),
+                                 _fbuf(C, np.float64)), "mx_dgemm_summa")
+        return C
+
+    def stats(self):
+        st = MxStats()
+        _ck(lib().mx_stats(self._ctx, ctypes.byref(st)))
+        return {f: getattr(st, f) for f, _ in MxStats._fields_}
+
+    # -- device-resident bench path ------------------------------------------
+    def alloc(self, nbytes):
+        b = ctypes.c_void_p()
+        _ck(lib().mx_alloc(self._ctx, nbytes, ctypes.byref(b)), "mx_alloc")
+        return b
+
+    def free(self, b):
+        _ck(lib().mx_free(self._ctx, b))
+
+    def fill_random(self, dbuf, n_elems, seed, fp32=False):
+        _ck(lib().mx_fill_random(self._ctx, dbuf, n_elems, seed,
+                                 1 if fp32 else 0), "mx_fill_random")
+
+    def dgemm_device(self, m, k, n, dA, lda, dB, ldb, dC, ldc):
+        _ck(lib().mx_dgemm_device(self._ctx, m, k, n, dA, lda, dB, ldb, dC,
+                                  ldc), "mx_dgemm_device")
+
+    def sgemm_device(self, m, k, n, dA, lda, dB, ldb, dC, ldc):
+        _ck(lib().mx_sgemm_device(self._ctx, m, k, n, dA, lda, dB, ldb, dC,
+                                  ldc), "mx_sgemm_device")
+
+    def dgemm_summa_device(self, m, k, n, dA, dB, dC):
+        _ck(lib().mx_dgemm_summa_device(self._ctx, m, k, n, dA, dB, dC),
+            "mx_dgemm_summa_device")
+
+    def download(self, host_arr, dbuf, nbytes):
+        _ck(lib().mx_download(self._ctx,
+                              host_arr.ctypes.data_as(ctypes.c_void_p), dbuf,
+                              nbytes), "mx_download")
+
+    def upload(self, dbuf, host_arr, nbytes):
+        _ck(lib().mx_upload(self._ctx, dbuf,
+                            host_arr.ctypes.data_as(ctypes.c_void_p), nbytes),
+            "mx_upload")

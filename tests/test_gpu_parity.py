@@ -1,0 +1,186 @@
+# GPU parity tests — the MFMA engine against the oracle restatement of the
+# reference's arithmetic, through the C ABI. Bar (BASELINE.json):
+# <= 1e-10 relative for fp64, <= 1e-4 for fp32. Run via:
+#   gpurun -- 'python -m pytest tests -m gpu -x -q'
+import ctypes
+import os
+
+import numpy as np
+import pytest
+
+from marlin_amd import Engine, EngineUnavailable, DenseVecMatrix
+from marlin_amd import engine as E
+from oracle import gen_matrix, blocked_multiply
+
+pytestmark = pytest.mark.gpu
+
+GOLD = os.path.join(os.path.dirname(os.path.abspath(__file__)), "golden")
+
+
+def rel_err(got, ref):
+    return np.max(np.abs(got - ref)) / max(np.max(np.abs(ref)), 1e-300)
+
+
+@pytest.fixture(scope="module")
+def eng():
+    e = Engine(0)
+    yield e
+    e.close()
+
+
+def test_native_extension_is_loaded():
+    # the product path must run our in-tree .so, not any fallback
+    assert os.path.exists(os.path.join(os.path.dirname(GOLD), "..",
+                                       "marlin_amd", "libmarlin_gpu.so"))
+    E.lib()  # raises if not loadable
+
+
+def test_golden_100x100(eng):
+    a = np.load(os.path.join(GOLD, "a100.npy"))
+    b = np.load(os.path.join(GOLD, "b100.npy"))
+    c_exp = np.load(os.path.join(GOLD, "c100.npy"))
+    assert rel_err(eng.dgemm(a, b), c_exp) < 1e-10
+
+
+def test_golden_4x4_routes(eng):
+    m4 = np.array([[0., 1, 2, 3], [2, 3, 4, 5], [3, 2, 1, 0], [1, 1, 1, 1]])
+    c4 = np.array([[11., 10, 9, 8], [23, 24, 25, 26], [7, 11, 15, 19],
+                   [6, 7, 8, 9]])
+    assert rel_err(eng.dgemm(m4, m4), c4) == 0.0
+    # operator surface: DenseVecMatrix route + BlockMatrix tile route
+    dvm = DenseVecMatrix(m4, engine=eng)
+    assert rel_err(dvm.multiply(m4).toBreeze(), c4) == 0.0
+    blk = dvm.toBlockMatrix(2, 2)
+    got = blk.multiply(blk)
+    assert rel_err(got.toBreeze(), c4) == 0.0
+    np.testing.assert_array_equal(got._blocks[(0, 0)],
+                                  [[11.0, 10.0], [23.0, 24.0]])
+
+
+@pytest.mark.parametrize("mkn", [
+    (128, 128, 128),          # exact single tile
+    (256, 512, 384),          # multi-tile aligned
+    (100, 100, 100),          # config 1 shape
+    (516, 300, 407),          # ragged everything
+    (1023, 517, 769),         # ragged odd
+    (1, 1000, 1),             # degenerate vector-ish
+    (129, 16, 127),           # barely over one tile
+])
+def test_dgemm_parity_vs_oracle(eng, mkn):
+    m, k, n = mkn
+    a = gen_matrix(m, k, seed=0xA11CE)
+    b = gen_matrix(k, n, seed=0xB0B)
+    ref = a @ b
+    assert rel_err(eng.dgemm(a, b), ref) < 1e-10
+
+
+def test_dgemm_matches_blocked_oracle_route(eng):
+    # the reference's own blocked route (ceil blocks, ascending-l reduce)
+    a = gen_matrix(700, 450, seed=7)
+    b = gen_matrix(450, 333, seed=8)
+    ref = blocked_multiply(a, b, (2, 2, 2))
+    assert rel_err(eng.dgemm(a, b), ref) < 1e-10
+
+
+def test_tile_dgemm_acc_combiner(eng):
+    # SubMatrix.multiply + add chain: C = sum_l A_l B_l
+    rng = [gen_matrix(96, 64, seed=s) for s in (1, 2, 3)]
+    rng2 = [gen_matrix(64, 80, seed=s) for s in (4, 5, 6)]
+    c = None
+    for a, b in zip(rng, rng2):
+        c = eng.tile_dgemm_acc(a, b, c)
+    ref = sum(a @ b for a, b in zip(rng, rng2))
+    assert rel_err(c, ref) < 1e-10
+
+
+def test_sgemm_parity(eng):
+    a = gen_matrix(516, 1024, seed=11, dtype=np.float32)
+    b = gen_matrix(1024, 300, seed=12, dtype=np.float32)
+    ref = a.astype(np.float64) @ b.astype(np.float64)
+    assert rel_err(eng.sgemm(a, b).astype(np.float64), ref) < 1e-4
+
+
+def test_sgemm_transpose_add_epilogue(eng):
+    a = gen_matrix(256, 384, seed=21, dtype=np.float32)
+    b = gen_matrix(384, 128, seed=22, dtype=np.float32)
+    add = gen_matrix(128, 256, seed=23, dtype=np.float32)
+    got = eng.sgemm_transpose_add(a, b, add)
+    ref = (a.astype(np.float64) @ b.astype(np.float64)).T + add
+    assert got.shape == (128, 256)
+    assert rel_err(got.astype(np.float64), ref) < 1e-4
+
+
+def test_fill_random_matches_oracle_bitexact(eng):
+    n = 1000
+    d = eng.alloc(n * 8)
+    eng.fill_random(d, n, seed=0xA11CE)
+    host = np.empty(n, dtype=np.float64)
+    eng.download(host, d, n * 8)
+    eng.free(d)
+    ref = gen_matrix(n, 1, seed=0xA11CE)[:, 0]
+    np.testing.assert_array_equal(host, ref)
+
+
+def test_summa_single_rank(eng):
+    # the SUMMA path end-to-end with a 1x1 grid (RCCL comm of size 1):
+    # same code path the 8-GPU run takes, minus inter-rank traffic
+    eng.comm_init(0, 1, Engine.comm_id())
+    m, k, n = 700, 9000, 300
+    a = gen_matrix(m, k, seed=31)
+    b = gen_matrix(k, n, seed=32)
+    got = eng.dgemm_summa(m, k, n, a, b)
+    assert rel_err(got, a @ b) < 1e-10
+
+
+def test_dimension_mismatch_error(eng):
+    a = gen_matrix(10, 11, seed=1)
+    b = gen_matrix(12, 10, seed=2)
+    with pytest.raises(ValueError):
+        eng.dgemm(a, b)
+
+
+def test_full_size_property_20000(eng):
+    # BASELINE config 3 size on one GPU, checked by a size-independent
+    # property (the oracle cannot do 20000^3 on CPU in test time):
+    # (A @ B) v == A (B v) for seeded v, elementwise to 1e-10 rel scale.
+    m = k = n = 20000
+    mp, kp, np_ = 20096, 20000, 20096
+    elem = 8
+    dA = eng.alloc(mp * kp * elem)
+    dB = eng.alloc(kp * np_ * elem)
+    dC = eng.alloc(mp * np_ * elem)
+    try:
+        eng.fill_random(dA, mp * kp, 0xA11CE)
+        eng.fill_random(dB, kp * np_, 0xB0B)
+        # zero the pad rows of A via re-fill of a clean padded layout:
+        # fill wrote random into pads too; pads only pollute C pad rows,
+        # which the property check below never reads.
+        eng.dgemm_device(mp, kp, np_, dA, mp, dB, kp, dC, mp)
+        # download row 0 and column 0 of C, plus A row 0, B col 0 strips
+        C0 = np.empty(np_, dtype=np.float64)      # C[0, :] strided
+        lib = E.lib()
+        # pull full C row 0: stride mp -> use 2D download via host loop is
+        # too slow; instead pull the first column (contiguous) and check
+        # C[:,0] == A @ B[:,0]
+        col = np.empty(mp, dtype=np.float64)
+        eng.download(col, dC, mp * 8)             # first column of C
+        bcol = np.empty(kp, dtype=np.float64)
+        eng.download(bcol, dB, kp * 8)            # first column of B
+        # host-side A @ b0 without materialising A on host: recompute A's
+        # entries streamingly from the generator in chunks
+        from oracle import gen_uniform_u64
+        acc = np.zeros(m, dtype=np.float64)
+        chunk = 512
+        for c0 in range(0, k, chunk):
+            c1 = min(c0 + chunk, k)
+            # A columns c0:c1 (padded pitch mp, random pads ignored)
+            z = gen_uniform_u64(0xA11CE, c0 * mp, (c1 - c0) * mp)
+            blockA = ((z >> np.uint64(11)).astype(np.float64) *
+                      2.0 ** -53).reshape((c1 - c0, mp)).T[:m]
+            acc += blockA @ bcol[c0:c1]
+        rel = np.max(np.abs(col[:m] - acc)) / np.max(np.abs(acc))
+        assert rel < 1e-10, rel
+    finally:
+        eng.free(dA)
+        eng.free(dB)
+        eng.free(dC)

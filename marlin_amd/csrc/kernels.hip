@@ -34,7 +34,9 @@ typedef float v4f __attribute__((ext_vector_type(4)));
 // MFMA wrappers (16x16x4 shape; one A/B element per lane):
 //   a: A[row = lane&15][k = lane>>4]      (A is m x k)
 //   b: B[k = lane>>4][col = lane&15]      (B is k x n)
-//   acc (4 per lane): D[row = (lane>>4)*4 + j][col = lane&15]
+//   acc (4 per lane): D[row = 4*j + (lane>>4)][col = lane&15]
+//     (measured on MI355X — tools_dev/debug_mfma.py; the f64/f32 16x16x4
+//      "SGEMM-class" row map is reg-major, unlike the bf16 16x16x32 map)
 DEVFN v4d mfma_16x16x4(double a, double b, v4d c) {
     return __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c, 0, 0, 0);
 }
@@ -129,17 +131,22 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
         }
     };
 
+    // C/D row of accumulator element j for this lane (measured per type:
+    // tools_dev/debug_mfma.py): f64 16x16x4 is reg-major (l4 + 4j).
+    constexpr bool REGMAJOR = sizeof(T) == 8;
+    const int rbase = REGMAJOR ? l4 : l4 * 4;
+    constexpr int rstep = REGMAJOR ? 4 : 1;
     ACC acc[4][4];
     if (BETA) {
         #pragma unroll
         for (int mt = 0; mt < 4; mt++)
             #pragma unroll
             for (int nt = 0; nt < 4; nt++) {
-                int64_t r = row0 + wm * 64 + mt * 16 + l4 * 4;
+                int64_t r = row0 + wm * 64 + mt * 16 + rbase;
                 int64_t cc = col0 + wn * 64 + nt * 16 + l15;
                 const T* cp = C + cc * ldc + r;
                 #pragma unroll
-                for (int j = 0; j < 4; j++) acc[mt][nt][j] = cp[j];
+                for (int j = 0; j < 4; j++) acc[mt][nt][j] = cp[rstep * j];
             }
     } else {
         #pragma unroll
@@ -184,16 +191,16 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
         __builtin_amdgcn_s_barrier();        // readers done before overwrite
     }
 
-    // --- epilogue: 4 contiguous rows per lane per fragment (col-major) ---
+    // --- epilogue (row map per type, see above) --------------------------
     #pragma unroll
     for (int mt = 0; mt < 4; mt++)
         #pragma unroll
         for (int nt = 0; nt < 4; nt++) {
-            int64_t r = row0 + wm * 64 + mt * 16 + l4 * 4;
+            int64_t r = row0 + wm * 64 + mt * 16 + rbase;
             int64_t cc = col0 + wn * 64 + nt * 16 + l15;
             T* cp = C + cc * ldc + r;
             #pragma unroll
-            for (int j = 0; j < 4; j++) cp[j] = acc[mt][nt][j];
+            for (int j = 0; j < 4; j++) cp[rstep * j] = acc[mt][nt][j];
         }
 }
 
@@ -294,8 +301,7 @@ __global__ void sgemm_mfma_tn_epilogue_kernel(
     }
 
     // transposed store: result element (r, c) of A*B goes to C[c + r*ldc];
-    // lane's 4 elements share column c -> contiguous in C_out? No: they are
-    // 4 consecutive r -> stride ldc apart. addC fused here.
+    // f32 16x16x4 row map: element j at row 4*l4 + j. addC fused.
     #pragma unroll
     for (int mt = 0; mt < 4; mt++)
         #pragma unroll

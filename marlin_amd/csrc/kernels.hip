@@ -108,6 +108,16 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
     constexpr int B_COLS_PER_GLDS = (64 * E) / BK;     // 8 (f64), 16 (f32)
     constexpr int B_GLDS = BN / B_COLS_PER_GLDS / 4;   // per wave: 4 (f64), 2 (f32)
 
+    // --- LDS bank-conflict swizzles (fp64 only; measured 8-way on the
+    // B-fragment reads without them — profiles/r01 PMC: conflict cycles
+    // were 88% of LDS cycles). glds forces a lane-linear LDS image, so the
+    // swizzle is applied to the per-lane GLOBAL source address (guide
+    // idiom); both XORs permute 16-byte chunks within one 128-byte global
+    // run, so HBM coalescing is unchanged.
+    //   A image: column kk stores row-pair i at chunk i ^ (8*(kk&1))
+    //     -> a-read banks: lanes 0-15 distinct, lanes 16-31 shifted by 32.
+    //   B image: column c stores k-pair p at chunk p ^ ((c>>1)&7)
+    //     -> b-read banks: all 32 lanes of a ds_read_b64 group distinct.
     auto issue_tile = [&](int kt, int buf) {
         const int64_t kbase = (int64_t)kt * BK;
         // A: column c of the tile = global column kbase+c; within a column
@@ -116,7 +126,11 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
         for (int i = 0; i < A_GLDS; i++) {
             int c = (wid * A_GLDS + i) * A_COLS_PER_GLDS;
             int lane_col = lane / (BM / E);            // 0 or extra col
-            int lane_row = (lane % (BM / E)) * E;
+            int lane_row;
+            if constexpr (E == 2)
+                lane_row = (lane ^ (((c + lane_col) & 1) << 3)) * E;
+            else
+                lane_row = (lane % (BM / E)) * E;
             const T* g = A + (kbase + c + lane_col) * lda + row0 + lane_row;
             glds16(g, &As[buf * BK * BM + c * BM]);
         }
@@ -125,7 +139,11 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
         for (int i = 0; i < B_GLDS; i++) {
             int c = (wid * B_GLDS + i) * B_COLS_PER_GLDS;
             int lane_col = lane / (BK / E);
-            int lane_row = (lane % (BK / E)) * E;
+            int lane_row;
+            if constexpr (E == 2)
+                lane_row = ((lane % 8) ^ (((c + lane_col) >> 1) & 7)) * E;
+            else
+                lane_row = (lane % (BK / E)) * E;
             const T* g = B + (col0 + c + lane_col) * ldb + kbase + lane_row;
             glds16(g, &Bs[buf * BN * BK + c * BK]);
         }
@@ -176,11 +194,23 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
             const int kk = q * 4 + l4;
             T a[4], b[4];
             #pragma unroll
-            for (int mt = 0; mt < 4; mt++)
-                a[mt] = At[kk * BM + wm * 64 + mt * 16 + l15];
+            for (int mt = 0; mt < 4; mt++) {
+                int rr = wm * 64 + mt * 16 + l15;
+                if constexpr (E == 2)
+                    a[mt] = At[kk * BM + (((rr >> 1) ^ ((kk & 1) << 3)) << 1)
+                               + (rr & 1)];
+                else
+                    a[mt] = At[kk * BM + rr];
+            }
             #pragma unroll
-            for (int nt = 0; nt < 4; nt++)
-                b[nt] = Bt[(wn * 64 + nt * 16 + l15) * BK + kk];
+            for (int nt = 0; nt < 4; nt++) {
+                int cb = wn * 64 + nt * 16 + l15;
+                if constexpr (E == 2)
+                    b[nt] = Bt[cb * BK + (((kk >> 1) ^ ((cb >> 1) & 7)) << 1)
+                               + (kk & 1)];
+                else
+                    b[nt] = Bt[cb * BK + kk];
+            }
             #pragma unroll
             for (int mt = 0; mt < 4; mt++)
                 #pragma unroll

@@ -39,11 +39,18 @@ def cpu_baseline_leg(args):
     per-tile dgemm) timed on the host cores — BASELINE.md plan. Bounded
     sample: nb^3 multiply (~seconds of CPU), scaled to TFLOP/s."""
     from oracle import gen_matrix, blocked_multiply, split_method
-    cores = os.cpu_count() or 1
+    ncpu = os.cpu_count() or 1
+    try:  # report the BLAS threads actually used, not just nproc
+        import threadpoolctl
+        infos = [i for i in threadpoolctl.threadpool_info()
+                 if i.get("user_api") == "blas"]
+        cores = max((i["num_threads"] for i in infos), default=1)
+    except Exception:
+        cores = ncpu
     nb = int(args.cpu_sample)
     a = gen_matrix(nb, nb, seed=0xA11CE)
     b = gen_matrix(nb, nb, seed=0xB0B)
-    mkn = split_method(nb, nb, nb, cores)
+    mkn = split_method(nb, nb, nb, ncpu)
     t0 = time.perf_counter()
     blocked_multiply(a, b, mkn)
     dt = time.perf_counter() - t0
@@ -51,8 +58,9 @@ def cpu_baseline_leg(args):
     return {
         "value": round(tf, 4), "unit": "TFLOP/s", "cores": cores,
         "kind": "port",
-        "sample": f"{nb}^3 fp64 blocked multiply (CARMA split "
-                  f"{mkn}, OpenBLAS tiles, {dt:.2f}s)",
+        "sample": f"{nb}^3 fp64 blocked multiply (CARMA split {mkn} for "
+                  f"{ncpu} host cores, OpenBLAS tiles x{cores} threads, "
+                  f"{dt:.2f}s)",
     }
 
 
@@ -67,7 +75,7 @@ def main():
     p.add_argument("--k", type=int, default=0)
     p.add_argument("--nn", type=int, default=0)
     p.add_argument("--dtype", choices=["f64", "f32"], default="f64")
-    p.add_argument("--cpu-sample", type=int, default=2048)
+    p.add_argument("--cpu-sample", type=int, default=4096)
     p.add_argument("--no-cpu-baseline", action="store_true")
     args = p.parse_args()
 

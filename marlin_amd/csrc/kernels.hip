@@ -24,6 +24,7 @@
 
 #include <hip/hip_runtime.h>
 #include <stdint.h>
+#include <stdlib.h>
 
 typedef double v4d __attribute__((ext_vector_type(4)));
 typedef float v4f __attribute__((ext_vector_type(4)));
@@ -57,38 +58,41 @@ DEVFN void glds16(const void* g, void* lds) {
 
 // ---------------------------------------------------------------------------
 // Main GEMM kernel. C[MxN] (+)= A[MxK] * B[KxN], col-major, padded pitches:
-// lda % 128 == 0 is NOT required (lda is the padded M), but M,N must be
-// multiples of 128 and K a multiple of 16 (host guarantees via padding).
-// BETA == 1 accumulates into C (the SubMatrix.add combiner, folded).
-template <typename T, int BETA>
-__launch_bounds__(256, 2)
+// M,N multiples of 128, K a multiple of BK (host pads to 32 -> both BK
+// configs divide it). BETA == 1 accumulates into C (the SubMatrix.add
+// combiner, folded into the MFMA accumulator chain).
+//
+// Geometry template: BM=BN=128 fixed; BK in {16, 32}; NWAVES in {4, 8}
+// (wave grid 2 x NWAVES/2). BK=32/NWAVES=8 = one 512-thread block per CU
+// (128 KB LDS), half the barrier rate of BK=16/NWAVES=4 (two 256-thread
+// blocks per CU, 64 KB LDS each); both keep 2 waves/SIMD.
+template <typename T, int BETA, int BK, int NWAVES>
+__launch_bounds__(NWAVES * 64, 2)
 __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
                                  const T* __restrict__ A, int64_t lda,
                                  const T* __restrict__ B, int64_t ldb,
                                  T* __restrict__ C, int64_t ldc,
                                  int nbm /* grid rows = M/128 */,
                                  int band /* column-band width in blocks */) {
-    constexpr int BM = 128, BN = 128, BK = 16;
+    constexpr int BM = 128, BN = 128;
     constexpr int E = 16 / sizeof(T);       // elems per 16B glds chunk
+    constexpr int WN = NWAVES / 2;          // wave grid 2 x WN
+    constexpr int MT = (BM / 2) / 16;       // 16x16 frags per wave, m
+    constexpr int NT = (BN / WN) / 16;      // 16x16 frags per wave, n
     using ACC = typename acc_t<T>::type;
 
     // --- block remap: column bands of `band` block-cols, bm fastest ------
-    int nbn = (int)(N / BN);
     int id = blockIdx.x;
     int per_band = nbm * band;
     int b0 = id / per_band;                  // band index
     int w = id - b0 * per_band;
-    int bn, bm;
-    int first_bn = b0 * band;
-    int bw = min(band, nbn - first_bn);      // last band may be narrower
-    if (bw == band) { bn = first_bn + w / nbm; bm = w % nbm; }
-    else            { bn = first_bn + w / nbm; bm = w % nbm; }
-    (void)bw;
+    int bn = b0 * band + w / nbm;
+    int bm = w % nbm;
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
-    const int wid = tid >> 6;                // 0..3
-    const int wm = wid >> 1, wn = wid & 1;   // 2x2 wave grid
+    const int wid = tid >> 6;
+    const int wm = wid / WN, wn = wid % WN;  // wave grid coords
     const int l15 = lane & 15, l4 = lane >> 4;   // MFMA fragment coords
 
     // --- LDS: A as [BK][BM] (k-col major), B as [BN][BK] (n-col major) ---
@@ -99,14 +103,14 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
     const int64_t row0 = (int64_t)bm * BM;    // global row of tile
     const int64_t col0 = (int64_t)bn * BN;    // global col of tile
 
-    // --- glds issue for one K-tile (kt) into buffer `buf` ----------------
-    // A tile: BK columns of BM elems; one glds fills BM*E... per glds a
-    // wave covers 64*E elems = (64*E)/BM columns. Waves split the BK cols.
-    constexpr int A_COLS_PER_GLDS = (64 * E) / BM;     // 1 (f64), 2 (f32)
-    constexpr int A_GLDS = BK / A_COLS_PER_GLDS / 4;   // per wave: 4 (f64), 2 (f32)
-    // B tile: BN columns of BK elems; one glds covers 64*E/BK columns.
-    constexpr int B_COLS_PER_GLDS = (64 * E) / BK;     // 8 (f64), 16 (f32)
-    constexpr int B_GLDS = BN / B_COLS_PER_GLDS / 4;   // per wave: 4 (f64), 2 (f32)
+    constexpr int A_COLS_PER_GLDS = (64 * E) / BM;
+    constexpr int A_GLDS = BK / A_COLS_PER_GLDS / NWAVES;
+    constexpr int B_COLS_PER_GLDS = (64 * E) / BK;
+    constexpr int B_GLDS = BN / B_COLS_PER_GLDS / NWAVES;
+    // B swizzle parameters: chunk index within a column is XORed with
+    // (c >> BSH) & BMASK (fp64 only; see below)
+    constexpr int BMASK = BK / 2 - 1;
+    constexpr int BSH = (BK == 16) ? 1 : 0;
 
     // --- LDS bank-conflict swizzles (fp64 only; measured 8-way on the
     // B-fragment reads without them — profiles/r01 PMC: conflict cycles
@@ -116,12 +120,11 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
     // run, so HBM coalescing is unchanged.
     //   A image: column kk stores row-pair i at chunk i ^ (8*(kk&1))
     //     -> a-read banks: lanes 0-15 distinct, lanes 16-31 shifted by 32.
-    //   B image: column c stores k-pair p at chunk p ^ ((c>>1)&7)
-    //     -> b-read banks: all 32 lanes of a ds_read_b64 group distinct.
+    //   B image: column c stores k-pair p at chunk p ^ ((c>>BSH)&BMASK)
+    //     -> b-read banks: all 32 lanes of a ds_read_b64 group distinct
+    //        (BK=16: column stride 128B; BK=32: column stride 256B).
     auto issue_tile = [&](int kt, int buf) {
         const int64_t kbase = (int64_t)kt * BK;
-        // A: column c of the tile = global column kbase+c; within a column
-        // lane covers rows E*(lane % (BM/E)); A_COLS_PER_GLDS cols per glds.
         #pragma unroll
         for (int i = 0; i < A_GLDS; i++) {
             int c = (wid * A_GLDS + i) * A_COLS_PER_GLDS;
@@ -134,14 +137,13 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
             const T* g = A + (kbase + c + lane_col) * lda + row0 + lane_row;
             glds16(g, &As[buf * BK * BM + c * BM]);
         }
-        // B: tile column c (global col0+c) holds BK k-elems contiguously.
         #pragma unroll
         for (int i = 0; i < B_GLDS; i++) {
             int c = (wid * B_GLDS + i) * B_COLS_PER_GLDS;
             int lane_col = lane / (BK / E);
             int lane_row;
             if constexpr (E == 2)
-                lane_row = ((lane % 8) ^ (((c + lane_col) >> 1) & 7)) * E;
+                lane_row = ((lane % (BK / E)) ^ (((c + lane_col) >> BSH) & BMASK)) * E;
             else
                 lane_row = (lane % (BK / E)) * E;
             const T* g = B + (col0 + c + lane_col) * ldb + kbase + lane_row;
@@ -150,27 +152,28 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
     };
 
     // C/D row of accumulator element j for this lane (measured per type:
-    // tools_dev/debug_mfma.py): f64 16x16x4 is reg-major (l4 + 4j).
+    // tools_dev/debug_mfma.py): f64 16x16x4 is reg-major (l4 + 4j),
+    // f32 16x16x4 is contiguous (4*l4 + j).
     constexpr bool REGMAJOR = sizeof(T) == 8;
     const int rbase = REGMAJOR ? l4 : l4 * 4;
     constexpr int rstep = REGMAJOR ? 4 : 1;
-    ACC acc[4][4];
+    ACC acc[MT][NT];
     if (BETA) {
         #pragma unroll
-        for (int mt = 0; mt < 4; mt++)
+        for (int mt = 0; mt < MT; mt++)
             #pragma unroll
-            for (int nt = 0; nt < 4; nt++) {
-                int64_t r = row0 + wm * 64 + mt * 16 + rbase;
-                int64_t cc = col0 + wn * 64 + nt * 16 + l15;
+            for (int nt = 0; nt < NT; nt++) {
+                int64_t r = row0 + wm * (MT * 16) + mt * 16 + rbase;
+                int64_t cc = col0 + wn * (NT * 16) + nt * 16 + l15;
                 const T* cp = C + cc * ldc + r;
                 #pragma unroll
                 for (int j = 0; j < 4; j++) acc[mt][nt][j] = cp[rstep * j];
             }
     } else {
         #pragma unroll
-        for (int mt = 0; mt < 4; mt++)
+        for (int mt = 0; mt < MT; mt++)
             #pragma unroll
-            for (int nt = 0; nt < 4; nt++) acc[mt][nt] = ACC{0};
+            for (int nt = 0; nt < NT; nt++) acc[mt][nt] = ACC{0};
     }
 
     const int ntiles = (int)(K / BK);
@@ -192,10 +195,10 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
         #pragma unroll
         for (int q = 0; q < BK / 4; q++) {
             const int kk = q * 4 + l4;
-            T a[4], b[4];
+            T a[MT], b[NT];
             #pragma unroll
-            for (int mt = 0; mt < 4; mt++) {
-                int rr = wm * 64 + mt * 16 + l15;
+            for (int mt = 0; mt < MT; mt++) {
+                int rr = wm * (MT * 16) + mt * 16 + l15;
                 if constexpr (E == 2)
                     a[mt] = At[kk * BM + (((rr >> 1) ^ ((kk & 1) << 3)) << 1)
                                + (rr & 1)];
@@ -203,18 +206,19 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
                     a[mt] = At[kk * BM + rr];
             }
             #pragma unroll
-            for (int nt = 0; nt < 4; nt++) {
-                int cb = wn * 64 + nt * 16 + l15;
+            for (int nt = 0; nt < NT; nt++) {
+                int cb = wn * (NT * 16) + nt * 16 + l15;
                 if constexpr (E == 2)
-                    b[nt] = Bt[cb * BK + (((kk >> 1) ^ ((cb >> 1) & 7)) << 1)
+                    b[nt] = Bt[cb * BK
+                               + ((((kk >> 1) ^ ((cb >> BSH) & BMASK)) << 1))
                                + (kk & 1)];
                 else
                     b[nt] = Bt[cb * BK + kk];
             }
             #pragma unroll
-            for (int mt = 0; mt < 4; mt++)
+            for (int mt = 0; mt < MT; mt++)
                 #pragma unroll
-                for (int nt = 0; nt < 4; nt++)
+                for (int nt = 0; nt < NT; nt++)
                     acc[mt][nt] = mfma_16x16x4(a[mt], b[nt], acc[mt][nt]);
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -223,11 +227,11 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
 
     // --- epilogue (row map per type, see above) --------------------------
     #pragma unroll
-    for (int mt = 0; mt < 4; mt++)
+    for (int mt = 0; mt < MT; mt++)
         #pragma unroll
-        for (int nt = 0; nt < 4; nt++) {
-            int64_t r = row0 + wm * 64 + mt * 16 + rbase;
-            int64_t cc = col0 + wn * 64 + nt * 16 + l15;
+        for (int nt = 0; nt < NT; nt++) {
+            int64_t r = row0 + wm * (MT * 16) + mt * 16 + rbase;
+            int64_t cc = col0 + wn * (NT * 16) + nt * 16 + l15;
             T* cp = C + cc * ldc + r;
             #pragma unroll
             for (int j = 0; j < 4; j++) cp[rstep * j] = acc[mt][nt][j];
@@ -395,26 +399,24 @@ int mxk_gemm(int is_fp32, int beta_one,
     if (M % 128 || N % 128 || K % 16) return -4;
     int nbm = (int)(M / 128), nbn = (int)(N / 128);
     int band = nbn < 8 ? nbn : 8;
-    dim3 grid((unsigned)(nbm * nbn)), block(256);
+    dim3 grid((unsigned)(nbm * nbn));
+    // config: BK=16 / 2x256-thread blocks per CU is the measured winner
+    // (67.3 vs 58.6 TF at 20000^3); MARLIN_GEMM_CFG=bk32 selects the
+    // 512-thread BK=32 variant (kept for re-evaluation).
+    static const char* cfg = getenv("MARLIN_GEMM_CFG");
+    bool bk32 = (K % 32 == 0) && cfg && cfg[0] == 'b' && cfg[2] == '3';
+    #define LAUNCH(TY, BETA, BK, NW)                                       \
+        hipLaunchKernelGGL((gemm_mfma_kernel<TY, BETA, BK, NW>), grid,     \
+            dim3(NW * 64), 0, stream, M, N, K, (const TY*)A, lda,          \
+            (const TY*)B, ldb, (TY*)C, ldc, nbm, band)
     if (is_fp32) {
-        if (beta_one)
-            hipLaunchKernelGGL((gemm_mfma_kernel<float, 1>), grid, block, 0, stream,
-                M, N, K, (const float*)A, lda, (const float*)B, ldb,
-                (float*)C, ldc, nbm, band);
-        else
-            hipLaunchKernelGGL((gemm_mfma_kernel<float, 0>), grid, block, 0, stream,
-                M, N, K, (const float*)A, lda, (const float*)B, ldb,
-                (float*)C, ldc, nbm, band);
+        if (beta_one) { if (bk32) LAUNCH(float, 1, 32, 8); else LAUNCH(float, 1, 16, 4); }
+        else          { if (bk32) LAUNCH(float, 0, 32, 8); else LAUNCH(float, 0, 16, 4); }
     } else {
-        if (beta_one)
-            hipLaunchKernelGGL((gemm_mfma_kernel<double, 1>), grid, block, 0, stream,
-                M, N, K, (const double*)A, lda, (const double*)B, ldb,
-                (double*)C, ldc, nbm, band);
-        else
-            hipLaunchKernelGGL((gemm_mfma_kernel<double, 0>), grid, block, 0, stream,
-                M, N, K, (const double*)A, lda, (const double*)B, ldb,
-                (double*)C, ldc, nbm, band);
+        if (beta_one) { if (bk32) LAUNCH(double, 1, 32, 8); else LAUNCH(double, 1, 16, 4); }
+        else          { if (bk32) LAUNCH(double, 0, 32, 8); else LAUNCH(double, 0, 16, 4); }
     }
+    #undef LAUNCH
     return (int)hipGetLastError() == 0 ? 0 : -2;
 }
 

@@ -69,12 +69,19 @@ def main():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
-    p.add_argument("--n", type=int, default=20000,
+    # size flags also readable from env (MARLIN_BENCH_*): torchrun's arg
+    # parser steals abbreviated flags like --n from the script args, so
+    # under torchrun sizes are passed via env instead
+    env = os.environ
+    p.add_argument("--n", type=int,
+                   default=int(env.get("MARLIN_BENCH_N", "20000")),
                    help="square problem size (m=k=n)")
-    p.add_argument("--m", type=int, default=0)
-    p.add_argument("--k", type=int, default=0)
-    p.add_argument("--nn", type=int, default=0)
-    p.add_argument("--dtype", choices=["f64", "f32"], default="f64")
+    p.add_argument("--m", type=int, default=int(env.get("MARLIN_BENCH_M", "0")))
+    p.add_argument("--k", type=int, default=int(env.get("MARLIN_BENCH_K", "0")))
+    p.add_argument("--nn", type=int,
+                   default=int(env.get("MARLIN_BENCH_NN", "0")))
+    p.add_argument("--dtype", choices=["f64", "f32"],
+                   default=env.get("MARLIN_BENCH_DTYPE", "f64"))
     p.add_argument("--cpu-sample", type=int, default=4096)
     p.add_argument("--no-cpu-baseline", action="store_true")
     args = p.parse_args()

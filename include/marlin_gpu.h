@@ -140,6 +140,23 @@ int mx_dgemm_summa_device(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
                           const mx_dbuf* dA_local, const mx_dbuf* dB_local,
                           mx_dbuf* dC_local);
 
+/* ---- elementwise / reduction / transpose (BlockMatrix epilogue ops,
+ * BlockMatrix.scala:344-523; DenseVecMatrix.scala scalar ops) ---------- */
+#define MX_OP_ADD   0   /* C = A + B          (add(other))              */
+#define MX_OP_SUB   1   /* C = A - B          (subtract(other))         */
+#define MX_OP_EMUL  2   /* C = A .* B         (dotProduct(other))       */
+#define MX_OP_ADDS  3   /* C = A + s          (add(b))                  */
+#define MX_OP_SUBS  4   /* C = A - s          (subtract(b))             */
+#define MX_OP_RSUBS 5   /* C = s - A          (subtractBy(b))           */
+#define MX_OP_MULS  6   /* C = A * s          (multiply(b))             */
+#define MX_OP_DIVS  7   /* C = A / s          (divide(b))               */
+#define MX_OP_RDIVS 8   /* C = s / A          (divideBy(b))             */
+int mx_map(mx_ctx* ctx, int op, int is_fp32, int64_t n, const void* A,
+           const void* B /* null for scalar ops */, double scalar, void* C);
+int mx_sum(mx_ctx* ctx, int is_fp32, int64_t n, const void* A, double* out);
+int mx_transpose(mx_ctx* ctx, int is_fp32, int64_t m, int64_t n,
+                 const void* A, void* C /* n x m col-major */);
+
 /* ---- timing / stats (MTUtils.evaluate + RMMcompare.scala:47-51 analog) -- */
 typedef struct {
   double h2d_ms;          /* host->device copies of the last call           */

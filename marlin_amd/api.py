@@ -83,6 +83,12 @@ class DenseVecMatrix:
         the same device GEMM, so dispatch only shapes the result type:
         a local ndarray route returns DenseVecMatrix, block routes
         return BlockMatrix with the split the reference would pick."""
+        if isinstance(other, (int, float)):
+            # multiply(b: Double) scalar overload (reference test
+            # `mat.multiply(2)`, DistributedMatrixSuite.scala:194)
+            return DenseVecMatrix(self._engine().map_op("muls", self._a,
+                                                        scalar=other),
+                                  self._eng)
         if isinstance(other, np.ndarray):
             # multiply(local Breeze matrix) -> DenseVecMatrix
             if self.numCols() != other.shape[0]:
@@ -108,6 +114,58 @@ class DenseVecMatrix:
         else:
             mkn = split_method(m, k, n, cores or 8)
         return _to_block(c, mkn[0], mkn[2], self._eng)
+
+    # -- elementwise family (DenseVecMatrix add/subtract/multiply/divide,
+    #    dotProduct, sum; same names/overloads as the reference) ----------
+    def add(self, other):
+        return self._ew2("add", "adds", other)
+
+    def subtract(self, other):
+        return self._ew2("sub", "subs", other)
+
+    def subtractBy(self, b):
+        return DenseVecMatrix(self._engine().map_op("rsubs", self._a,
+                                                    scalar=b), self._eng)
+
+    def multiply_scalar(self, b):
+        return DenseVecMatrix(self._engine().map_op("muls", self._a,
+                                                    scalar=b), self._eng)
+
+    def divide(self, b):
+        return DenseVecMatrix(self._engine().map_op("divs", self._a,
+                                                    scalar=b), self._eng)
+
+    def divideBy(self, b):
+        return DenseVecMatrix(self._engine().map_op("rdivs", self._a,
+                                                    scalar=b), self._eng)
+
+    def dotProduct(self, other):
+        if isinstance(other, BlockMatrix):
+            other = other.toDenseVecMatrix()
+        if (self.numRows() != other.numRows()
+                or self.numCols() != other.numCols()):
+            raise ValueError("matrix dimension mismatch")
+        return DenseVecMatrix(self._engine().map_op("emul", self._a,
+                                                    other._a), self._eng)
+
+    def sum(self):
+        return self._engine().sum(self._a)
+
+    def transpose(self):
+        return DenseVecMatrix(self._engine().transpose(self._a), self._eng)
+
+    def _ew2(self, op2, op1, other):
+        if isinstance(other, (int, float)):
+            return DenseVecMatrix(self._engine().map_op(op1, self._a,
+                                                        scalar=other),
+                                  self._eng)
+        if isinstance(other, BlockMatrix):
+            other = other.toDenseVecMatrix()
+        if (self.numRows() != other.numRows()
+                or self.numCols() != other.numCols()):
+            raise ValueError("matrix dimension mismatch")
+        return DenseVecMatrix(self._engine().map_op(op2, self._a, other._a),
+                              self._eng)
 
     def toBlockMatrix(self, blks_by_row, blks_by_col):
         """DenseVecMatrix.toBlockMatrix (DenseVecMatrix.scala:1259-1328)."""
@@ -167,6 +225,16 @@ class BlockMatrix:
         re-shuffles (emit x n / x m, join, reduceByKey); the engine computes
         the same per-C-tile sums with one owner per tile, so only the
         per-tile GEMM-accumulate chain remains (mx_tile_dgemm_acc)."""
+        if isinstance(other, (int, float)):
+            return self._ew1("muls", other)
+        if isinstance(other, DenseVecMatrix):
+            if self.numCols() != other.numRows():
+                raise ValueError(
+                    f"Dimension mismatch during matrix-matrix multiplication: "
+                    f"{self.numCols()} vs {other.numRows()}")
+            # BlockMatrix x DenseVecMatrix (BlockMatrix.scala:305-335)
+            return DenseVecMatrix(
+                self._engine().dgemm(self.toBreeze(), other._a), self._eng)
         if self.numCols() != other.numRows():
             raise ValueError(
                 f"Dimension mismatch during matrix-matrix multiplication: "
@@ -188,6 +256,69 @@ class BlockMatrix:
                 out[(i, j)] = acc
         return BlockMatrix(out, self.numRows(), other.numCols(),
                            engine=self._eng)
+
+    # -- elementwise family (BlockMatrix.scala:344-523 names) ------------
+    def add(self, other):
+        return self._ew2("add", "adds", other)
+
+    def subtract(self, other):
+        return self._ew2("sub", "subs", other)
+
+    def subtractBy(self, b):
+        return self._ew1("rsubs", b)
+
+    def multiply_scalar(self, b):
+        return self._ew1("muls", b)
+
+    def divide(self, b):
+        return self._ew1("divs", b)
+
+    def divideBy(self, b):
+        return self._ew1("rdivs", b)
+
+    def dotProduct(self, other):
+        if isinstance(other, DenseVecMatrix):
+            other = _to_block(other._a, self._nbr, self._nbc, self._eng)
+        if (self.numRows() != other.numRows()
+                or self.numCols() != other.numCols()):
+            raise ValueError("matrix dimension mismatch")
+        eng = self._engine()
+        out = {k: eng.map_op("emul", v, other._blocks[k])
+               for k, v in self._blocks.items()}
+        return BlockMatrix(out, self._rows, self._cols, engine=self._eng)
+
+    def sum(self):
+        eng = self._engine()
+        return float(sum(eng.sum(v) for v in self._blocks.values()))
+
+    def transpose(self):
+        """BlockMatrix.transpose (BlockMatrix.scala:514-523): per-block
+        device transpose + block-ID swap."""
+        eng = self._engine()
+        out = {(j, i): eng.transpose(v) for (i, j), v in self._blocks.items()}
+        return BlockMatrix(out, self._cols, self._rows, engine=self._eng)
+
+    def _ew1(self, op, b):
+        eng = self._engine()
+        out = {k: eng.map_op(op, v, scalar=b) for k, v in self._blocks.items()}
+        return BlockMatrix(out, self._rows, self._cols, engine=self._eng)
+
+    def _ew2(self, op2, op1, other):
+        if isinstance(other, (int, float)):
+            return self._ew1(op1, other)
+        if isinstance(other, DenseVecMatrix):
+            other = _to_block(other._a, self._nbr, self._nbc, self._eng)
+        if (self.numRows() != other.numRows()
+                or self.numCols() != other.numCols()):
+            raise ValueError("matrix dimension mismatch")
+        if (self.numBlksByRow() != other.numBlksByRow()
+                or self.numBlksByCol() != other.numBlksByCol()):
+            other = _to_block(other.toBreeze(), self._nbr, self._nbc,
+                              self._eng)
+        eng = self._engine()
+        out = {k: eng.map_op(op2, v, other._blocks[k])
+               for k, v in self._blocks.items()}
+        return BlockMatrix(out, self._rows, self._cols, engine=self._eng)
 
     def toDenseVecMatrix(self):
         """BlockMatrix.toDenseVecMatrix (BlockMatrix.scala:575-594)."""

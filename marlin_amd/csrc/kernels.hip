@@ -389,6 +389,108 @@ __global__ void zero_pad_kernel(T* __restrict__ buf, int64_t rows_total,
 }
 
 // ---------------------------------------------------------------------------
+// Elementwise / scalar op family — the BlockMatrix epilogue ops
+// (BlockMatrix.scala:344-507: add/subtract matrix+scalar, subtractBy,
+// divide, divideBy, dotProduct=element-wise multiply; DenseVecMatrix
+// multiply(scalar)). HBM-bound map kernels; layout-agnostic (flat).
+enum MxMapOp {
+    MX_OP_ADD = 0,    // C = A + B
+    MX_OP_SUB = 1,    // C = A - B
+    MX_OP_EMUL = 2,   // C = A .* B   (reference "dotProduct")
+    MX_OP_ADDS = 3,   // C = A + s
+    MX_OP_SUBS = 4,   // C = A - s
+    MX_OP_RSUBS = 5,  // C = s - A    (subtractBy)
+    MX_OP_MULS = 6,   // C = A * s
+    MX_OP_DIVS = 7,   // C = A / s
+    MX_OP_RDIVS = 8,  // C = s / A    (divideBy)
+};
+
+template <typename T>
+__global__ void map_kernel(int op, int64_t n, const T* __restrict__ a,
+                           const T* __restrict__ b, double scalar,
+                           T* __restrict__ c) {
+    const T s = (T)scalar;
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        T x = a[i];
+        T y;
+        switch (op) {
+            case MX_OP_ADD:   y = x + b[i]; break;
+            case MX_OP_SUB:   y = x - b[i]; break;
+            case MX_OP_EMUL:  y = x * b[i]; break;
+            case MX_OP_ADDS:  y = x + s; break;
+            case MX_OP_SUBS:  y = x - s; break;
+            case MX_OP_RSUBS: y = s - x; break;
+            case MX_OP_MULS:  y = x * s; break;
+            case MX_OP_DIVS:  y = x / s; break;
+            default:          y = s / x; break;
+        }
+        c[i] = y;
+    }
+}
+
+// sum reduction (DistributedMatrix.sum, BlockMatrix.scala sum tests):
+// stage 1: per-block tree sums -> partials; stage 2: one block combines
+// in fixed order (deterministic given the fixed grid).
+template <typename T>
+__global__ void sum_stage1_kernel(int64_t n, const T* __restrict__ a,
+                                  double* __restrict__ partials) {
+    __shared__ double sm[256];
+    double acc = 0;
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) acc += (double)a[i];
+    sm[threadIdx.x] = acc;
+    __syncthreads();
+    for (int w = 128; w > 0; w >>= 1) {
+        if (threadIdx.x < w) sm[threadIdx.x] += sm[threadIdx.x + w];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) partials[blockIdx.x] = sm[0];
+}
+
+__global__ void sum_stage2_kernel(int nparts, const double* __restrict__ p,
+                                  double* __restrict__ out) {
+    __shared__ double sm[256];
+    double acc = 0;
+    for (int i = threadIdx.x; i < nparts; i += 256) acc += p[i];
+    sm[threadIdx.x] = acc;
+    __syncthreads();
+    for (int w = 128; w > 0; w >>= 1) {
+        if (threadIdx.x < w) sm[threadIdx.x] += sm[threadIdx.x + w];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) *out = sm[0];
+}
+
+// Tiled transpose (BlockMatrix.transpose, BlockMatrix.scala:514-523):
+// out[n x m] = in[m x n]^T, col-major, 32x32 LDS tiles (+1 pad), both
+// sides coalesced.
+template <typename T>
+__global__ void transpose_kernel(int64_t m, int64_t n,
+                                 const T* __restrict__ in,
+                                 T* __restrict__ out) {
+    __shared__ T tile[32][33];
+    int64_t tm = (m + 31) / 32;
+    int64_t bi = blockIdx.x % tm;           // tile row of input
+    int64_t bj = blockIdx.x / tm;           // tile col of input
+    int tx = threadIdx.x % 32, ty = threadIdx.x / 32;  // 32x8 threads
+    int64_t r0 = bi * 32, c0 = bj * 32;
+    #pragma unroll
+    for (int k = 0; k < 4; k++) {
+        int64_t r = r0 + tx, c = c0 + ty + k * 8;
+        if (r < m && c < n) tile[ty + k * 8][tx] = in[c * m + r];
+    }
+    __syncthreads();
+    #pragma unroll
+    for (int k = 0; k < 4; k++) {
+        int64_t r = c0 + tx, c = r0 + ty + k * 8;   // output coords
+        if (r < n && c < m) out[c * n + r] = tile[tx][ty + k * 8];
+    }
+}
+
+// ---------------------------------------------------------------------------
 // C-visible launchers (called from marlin_gpu.cpp).
 extern "C" {
 
@@ -455,6 +557,46 @@ int mxk_zero_pad(int is_fp32, void* buf, int64_t rows_total, int64_t cols_total,
     else
         hipLaunchKernelGGL(zero_pad_kernel<double>, grid, block, 0, stream,
                            (double*)buf, rows_total, cols_total, ld, m, n);
+    return (int)hipGetLastError() == 0 ? 0 : -2;
+}
+
+int mxk_map(int is_fp32, int op, int64_t n, const void* a, const void* b,
+            double scalar, void* c, hipStream_t stream) {
+    dim3 grid(2048), block(256);
+    if (is_fp32)
+        hipLaunchKernelGGL(map_kernel<float>, grid, block, 0, stream, op, n,
+                           (const float*)a, (const float*)b, scalar, (float*)c);
+    else
+        hipLaunchKernelGGL(map_kernel<double>, grid, block, 0, stream, op, n,
+                           (const double*)a, (const double*)b, scalar,
+                           (double*)c);
+    return (int)hipGetLastError() == 0 ? 0 : -2;
+}
+
+int mxk_sum(int is_fp32, int64_t n, const void* a, double* partials,
+            double* out, hipStream_t stream) {
+    dim3 grid(1024), block(256);
+    if (is_fp32)
+        hipLaunchKernelGGL(sum_stage1_kernel<float>, grid, block, 0, stream,
+                           n, (const float*)a, partials);
+    else
+        hipLaunchKernelGGL(sum_stage1_kernel<double>, grid, block, 0, stream,
+                           n, (const double*)a, partials);
+    hipLaunchKernelGGL(sum_stage2_kernel, dim3(1), block, 0, stream, 1024,
+                       partials, out);
+    return (int)hipGetLastError() == 0 ? 0 : -2;
+}
+
+int mxk_transpose(int is_fp32, int64_t m, int64_t n, const void* in,
+                  void* out, hipStream_t stream) {
+    int64_t tm = (m + 31) / 32, tn = (n + 31) / 32;
+    dim3 grid((unsigned)(tm * tn)), block(256);
+    if (is_fp32)
+        hipLaunchKernelGGL(transpose_kernel<float>, grid, block, 0, stream,
+                           m, n, (const float*)in, (float*)out);
+    else
+        hipLaunchKernelGGL(transpose_kernel<double>, grid, block, 0, stream,
+                           m, n, (const double*)in, (double*)out);
     return (int)hipGetLastError() == 0 ? 0 : -2;
 }
 

@@ -38,6 +38,12 @@ int mxk_fill_random(int is_fp32, void* buf, int64_t m, int64_t n, int64_t ld,
 int mxk_zero_pad(int is_fp32, void* buf, int64_t rows_total,
                  int64_t cols_total, int64_t ld, int64_t m, int64_t n,
                  hipStream_t stream);
+int mxk_map(int is_fp32, int op, int64_t n, const void* a, const void* b,
+            double scalar, void* c, hipStream_t stream);
+int mxk_sum(int is_fp32, int64_t n, const void* a, double* partials,
+            double* out, hipStream_t stream);
+int mxk_transpose(int is_fp32, int64_t m, int64_t n, const void* in,
+                  void* out, hipStream_t stream);
 }
 
 #define HIP_OK(x)                                                        \
@@ -589,6 +595,70 @@ int mx_dgemm_summa(mx_ctx* c, int64_t m, int64_t k, int64_t n,
 int mx_sgemm_summa(mx_ctx* c, int64_t m, int64_t k, int64_t n, const float* A,
                    const float* B, float* C) {
   return summa_host(c, 1, m, k, n, A, B, C);
+}
+
+// ---------------------------------------------------------------------------
+// Elementwise / reduction / transpose host entries (the BlockMatrix
+// epilogue-op family, BlockMatrix.scala:344-523). Host col-major buffers;
+// elementwise ops are layout-agnostic (flat n elements).
+int mx_map(mx_ctx* c, int op, int is_fp32, int64_t n, const void* A,
+           const void* B, double scalar, void* C) {
+  if (!c || !A || !C || n <= 0) return MX_EINVAL;
+  const int binary = op <= 2;
+  if (binary && !B) return MX_EINVAL;
+  const int64_t elem = is_fp32 ? 4 : 8;
+  HIP_OK(hipSetDevice(c->device));
+  int rc;
+  if ((rc = ensure(c, &c->wsA, n * elem))) return rc;
+  if (binary && (rc = ensure(c, &c->wsB, n * elem))) return rc;
+  HIP_OK(hipMemcpyAsync(c->wsA.ptr, A, n * elem, hipMemcpyHostToDevice,
+                        c->s_gemm));
+  if (binary)
+    HIP_OK(hipMemcpyAsync(c->wsB.ptr, B, n * elem, hipMemcpyHostToDevice,
+                          c->s_gemm));
+  if ((rc = mxk_map(is_fp32, op, n, c->wsA.ptr, binary ? c->wsB.ptr : nullptr,
+                    scalar, c->wsA.ptr, c->s_gemm)))
+    return MX_EHIP;
+  HIP_OK(hipMemcpyAsync(C, c->wsA.ptr, n * elem, hipMemcpyDeviceToHost,
+                        c->s_gemm));
+  HIP_OK(hipStreamSynchronize(c->s_gemm));
+  return MX_OK;
+}
+
+int mx_sum(mx_ctx* c, int is_fp32, int64_t n, const void* A, double* out) {
+  if (!c || !A || !out || n <= 0) return MX_EINVAL;
+  const int64_t elem = is_fp32 ? 4 : 8;
+  HIP_OK(hipSetDevice(c->device));
+  int rc;
+  if ((rc = ensure(c, &c->wsA, n * elem))) return rc;
+  if ((rc = ensure(c, &c->wsB, (1024 + 1) * 8))) return rc;
+  HIP_OK(hipMemcpyAsync(c->wsA.ptr, A, n * elem, hipMemcpyHostToDevice,
+                        c->s_gemm));
+  double* parts = (double*)c->wsB.ptr;
+  if ((rc = mxk_sum(is_fp32, n, c->wsA.ptr, parts, parts + 1024, c->s_gemm)))
+    return MX_EHIP;
+  HIP_OK(hipMemcpyAsync(out, parts + 1024, 8, hipMemcpyDeviceToHost,
+                        c->s_gemm));
+  HIP_OK(hipStreamSynchronize(c->s_gemm));
+  return MX_OK;
+}
+
+int mx_transpose(mx_ctx* c, int is_fp32, int64_t m, int64_t n, const void* A,
+                 void* C) {
+  if (!c || !A || !C || m <= 0 || n <= 0) return MX_EINVAL;
+  const int64_t elem = is_fp32 ? 4 : 8;
+  HIP_OK(hipSetDevice(c->device));
+  int rc;
+  if ((rc = ensure(c, &c->wsA, m * n * elem))) return rc;
+  if ((rc = ensure(c, &c->wsB, m * n * elem))) return rc;
+  HIP_OK(hipMemcpyAsync(c->wsA.ptr, A, m * n * elem, hipMemcpyHostToDevice,
+                        c->s_gemm));
+  if ((rc = mxk_transpose(is_fp32, m, n, c->wsA.ptr, c->wsB.ptr, c->s_gemm)))
+    return MX_EHIP;
+  HIP_OK(hipMemcpyAsync(C, c->wsB.ptr, m * n * elem, hipMemcpyDeviceToHost,
+                        c->s_gemm));
+  HIP_OK(hipStreamSynchronize(c->s_gemm));
+  return MX_OK;
 }
 
 int mx_stats(mx_ctx* c, mx_stats_t* out) {

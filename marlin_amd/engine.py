@@ -84,6 +84,10 @@ def _load():
         "mx_sgemm_device": (ctypes.c_int, [vp, i64, i64, i64, vp, i64, vp, i64,
                                            vp, i64]),
         "mx_dgemm_summa_device": (ctypes.c_int, [vp, i64, i64, i64, vp, vp, vp]),
+        "mx_map": (ctypes.c_int, [vp, ctypes.c_int, ctypes.c_int, i64, vp,
+                                  vp, dbl, vp]),
+        "mx_sum": (ctypes.c_int, [vp, ctypes.c_int, i64, vp, P(dbl)]),
+        "mx_transpose": (ctypes.c_int, [vp, ctypes.c_int, i64, i64, vp, vp]),
         "mx_stats": (ctypes.c_int, [vp, P(MxStats)]),
     }
     for name, (res, args) in sigs.items():
@@ -245,6 +249,45 @@ class Engine:
         _ck(lib().mx_dgemm_summa(self._ctx, m, k, n, _fbuf(A_local, np.float64),
                                  _fbuf(B_local, np.float64),
                                  _fbuf(C, np.float64)), "mx_dgemm_summa")
+        return C
+
+    # -- elementwise / reduction / transpose (epilogue-op family) ----------
+    OPS = {"add": 0, "sub": 1, "emul": 2, "adds": 3, "subs": 4, "rsubs": 5,
+           "muls": 6, "divs": 7, "rdivs": 8}
+
+    def map_op(self, op, A, B=None, scalar=0.0):
+        """Elementwise op on same-shape col-major arrays (flat)."""
+        A = np.asfortranarray(A, dtype=np.float64)
+        opc = self.OPS[op]
+        bp = None
+        if opc <= 2:
+            B = np.asfortranarray(B, dtype=np.float64)
+            if A.shape != B.shape:
+                raise ValueError("matrix dimension mismatch")
+            bp = B.ctypes.data_as(ctypes.c_void_p)
+        C = np.empty_like(A)
+        _ck(lib().mx_map(self._ctx, opc, 0, A.size,
+                         A.ctypes.data_as(ctypes.c_void_p), bp,
+                         float(scalar), C.ctypes.data_as(ctypes.c_void_p)),
+            "mx_map")
+        return C
+
+    def sum(self, A):
+        A = np.asfortranarray(A, dtype=np.float64)
+        out = ctypes.c_double()
+        _ck(lib().mx_sum(self._ctx, 0, A.size,
+                         A.ctypes.data_as(ctypes.c_void_p),
+                         ctypes.byref(out)), "mx_sum")
+        return out.value
+
+    def transpose(self, A):
+        A = np.asfortranarray(A, dtype=np.float64)
+        m, n = A.shape
+        C = np.empty((n, m), dtype=np.float64, order="F")
+        _ck(lib().mx_transpose(self._ctx, 0, m, n,
+                               A.ctypes.data_as(ctypes.c_void_p),
+                               C.ctypes.data_as(ctypes.c_void_p)),
+            "mx_transpose")
         return C
 
     def stats(self):

@@ -81,13 +81,32 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
     constexpr int NT = (BN / WN) / 16;      // 16x16 frags per wave, n
     using ACC = typename acc_t<T>::type;
 
-    // --- block remap: column bands of `band` block-cols, bm fastest ------
+    // --- block remap: column bands of `band` block-cols, split into 8x8
+    // block supertiles (band == 8 normally). Within a supertile the 8
+    // blocks landing on one XCD (dispatch places block b on XCD b%8, and
+    // all the stride terms are multiples of 8) form one COLUMN: each
+    // XCD's L2 re-reads one B tile 8x and a contiguous 8-row A panel.
+    // band < 0: plain bm-fastest bands (A/B comparison fallback).
     int id = blockIdx.x;
-    int per_band = nbm * band;
-    int b0 = id / per_band;                  // band index
-    int w = id - b0 * per_band;
-    int bn = b0 * band + w / nbm;
-    int bm = w % nbm;
+    int bn, bm;
+    {
+        int abands = band < 0 ? -band : band;
+        int nbn = (int)(N / BN);
+        int per_band = nbm * abands;
+        int b0 = id / per_band;
+        int w = id - b0 * per_band;
+        int first_bn = b0 * abands;
+        int bw = min(abands, nbn - first_bn);
+        if (band < 0) {                      // plain band order
+            bn = first_bn + w / nbm;
+            bm = w % nbm;
+        } else {                             // supertiled band order
+            int sh = 8 * bw;
+            int t = w / sh, l = w - t * sh;
+            bm = t * 8 + l / bw;
+            bn = first_bn + l % bw;
+        }
+    }
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -132,8 +151,8 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
             int lane_row;
             if constexpr (E == 2)
                 lane_row = (lane ^ (((c + lane_col) & 1) << 3)) * E;
-            else
-                lane_row = (lane % (BM / E)) * E;
+            else  // f32: 16B chunk (4 elems) XOR 4 by k parity (banks mod 32)
+                lane_row = ((lane % (BM / E)) ^ (((c + lane_col) & 1) << 2)) * E;
             const T* g = A + (kbase + c + lane_col) * lda + row0 + lane_row;
             glds16(g, &As[buf * BK * BM + c * BM]);
         }
@@ -144,6 +163,9 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
             int lane_row;
             if constexpr (E == 2)
                 lane_row = ((lane % (BK / E)) ^ (((c + lane_col) >> BSH) & BMASK)) * E;
+            else if constexpr (E == 4 && BK == 16)
+                // f32: 4 chunks/col; XOR by (c>>1)&3 -> 2-way worst case
+                lane_row = ((lane % 4) ^ (((c + lane_col) >> 1) & 3)) * E;
             else
                 lane_row = (lane % (BK / E)) * E;
             const T* g = B + (col0 + c + lane_col) * ldb + kbase + lane_row;
@@ -203,7 +225,8 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
                     a[mt] = At[kk * BM + (((rr >> 1) ^ ((kk & 1) << 3)) << 1)
                                + (rr & 1)];
                 else
-                    a[mt] = At[kk * BM + rr];
+                    a[mt] = At[kk * BM + (((rr >> 2) ^ ((kk & 1) << 2)) << 2)
+                               + (rr & 3)];
             }
             #pragma unroll
             for (int nt = 0; nt < NT; nt++) {
@@ -212,6 +235,10 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
                     b[nt] = Bt[cb * BK
                                + ((((kk >> 1) ^ ((cb >> BSH) & BMASK)) << 1))
                                + (kk & 1)];
+                else if constexpr (E == 4 && BK == 16)
+                    b[nt] = Bt[cb * BK
+                               + (((kk >> 2) ^ ((cb >> 1) & 3)) << 2)
+                               + (kk & 3)];
                 else
                     b[nt] = Bt[cb * BK + kk];
             }
@@ -501,6 +528,9 @@ int mxk_gemm(int is_fp32, int beta_one,
     if (M % 128 || N % 128 || K % 16) return -4;
     int nbm = (int)(M / 128), nbn = (int)(N / 128);
     int band = nbn < 8 ? nbn : 8;
+    // default: supertiled bands; MARLIN_GEMM_REMAP=bands -> plain bands
+    static const char* remap = getenv("MARLIN_GEMM_REMAP");
+    if (remap && remap[0] == 'b') band = -band;
     dim3 grid((unsigned)(nbm * nbn));
     // config: BK=16 / 2x256-thread blocks per CU is the measured winner
     // (67.3 vs 58.6 TF at 20000^3); MARLIN_GEMM_CFG=bk32 selects the

@@ -179,13 +179,26 @@ def main():
             ach = (2.0 * m * k * n * args.steps) / (gemm_ms_acc / 1e3) / 1e12
         else:
             ach = None
+        # HBM traffic per launch: measured offline by rocprofv3 --pmc
+        # FETCH_SIZE passes (tools_dev/rocpd_stats.py) and committed under
+        # profiles/hbm_traffic.json keyed by workload; null when this
+        # workload has no committed measurement.
+        traffic = None
+        try:
+            tj = json.load(open(os.path.join(HERE, "profiles",
+                                             "hbm_traffic.json")))
+            rec = tj.get(f"dgemm_{m}x{k}x{n}_{args.dtype}")
+            if rec and world == 1:
+                traffic = rec["reads_bytes_per_launch"]
+        except Exception:
+            pass
         roofline = {
             "bound": "mfma",
             "achieved": round(ach, 3) if ach else None,
             "peak": peak * world,
             "unit": "TFLOP/s",
             "frac": round(ach / (peak * world), 4) if ach else None,
-            "traffic": None,  # rocprofv3 PMC summaries: profiles/
+            "traffic": traffic,
         }
         out = {
             "metric": "dense C=AxB TFLOP/s (fp64)" if not fp32 else

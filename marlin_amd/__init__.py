@@ -3,3 +3,5 @@
 # behind the C ABI of include/marlin_gpu.h. No CPU fallback anywhere.
 from .engine import Engine, EngineError, EngineUnavailable  # noqa: F401
 from .api import DenseVecMatrix, BlockMatrix, BlockID, split_method  # noqa: F401
+from .io import (load_matrix_file, save_matrix_file,  # noqa: F401
+                 load_block_matrix_file, save_block_matrix_file)

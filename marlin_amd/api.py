@@ -171,6 +171,11 @@ class DenseVecMatrix:
         """DenseVecMatrix.toBlockMatrix (DenseVecMatrix.scala:1259-1328)."""
         return _to_block(self._a, blks_by_row, blks_by_col, self._eng)
 
+    def saveToFileSystem(self, path):
+        """DenseVecMatrix.saveToFileSystem (DenseVecMatrix.scala:1042)."""
+        from .io import save_matrix_file
+        save_matrix_file(self, path)
+
     def toBreeze(self):
         return self._a.copy()
 
@@ -319,6 +324,14 @@ class BlockMatrix:
         out = {k: eng.map_op(op2, v, other._blocks[k])
                for k, v in self._blocks.items()}
         return BlockMatrix(out, self._rows, self._cols, engine=self._eng)
+
+    def saveToFileSystem(self, path, format=" "):
+        """BlockMatrix.saveToFileSystem (BlockMatrix.scala:538-559)."""
+        from .io import save_block_matrix_file, save_matrix_file
+        if format.lower() == "blockmatrix":
+            save_block_matrix_file(self, path)
+        else:
+            save_matrix_file(self, path)
 
     def toDenseVecMatrix(self):
         """BlockMatrix.toDenseVecMatrix (BlockMatrix.scala:575-594)."""

@@ -41,7 +41,7 @@ def save_matrix_file(mat, path):
     a = mat.toBreeze()
     with open(path, "w") as f:
         for i in range(a.shape[0]):
-            f.write(f"{i}:" + ",".join(repr(v) for v in a[i]) + "\n")
+            f.write(f"{i}:" + ",".join(repr(float(v)) for v in a[i]) + "\n")
 
 
 def load_block_matrix_file(path, engine=None):
@@ -64,5 +64,5 @@ def save_block_matrix_file(mat, path):
     """BlockMatrix.saveToFileSystem(path, "blockmatrix")."""
     with open(path, "w") as f:
         for (r, c), blk in sorted(mat._blocks.items()):
-            data = ",".join(repr(v) for v in blk.flatten(order="F"))
+            data = ",".join(repr(float(v)) for v in blk.flatten(order="F"))
             f.write(f"{r}-{c}-{blk.shape[0]}-{blk.shape[1]}:{data}\n")

@@ -157,6 +157,11 @@ int mx_sum(mx_ctx* ctx, int is_fp32, int64_t n, const void* A, double* out);
 int mx_transpose(mx_ctx* ctx, int is_fp32, int64_t m, int64_t n,
                  const void* A, void* C /* n x m col-major */);
 
+/* Matrix-vector multiply (BlockMatrix.multiply(DistributedVector/BDV),
+ * BlockMatrix.scala:240-274): y = A x, A col-major m x n. */
+int mx_dgemv(mx_ctx* ctx, int64_t m, int64_t n, const double* A,
+             const double* x, double* y);
+
 /* ---- timing / stats (MTUtils.evaluate + RMMcompare.scala:47-51 analog) -- */
 typedef struct {
   double h2d_ms;          /* host->device copies of the last call           */

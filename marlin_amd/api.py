@@ -89,6 +89,9 @@ class DenseVecMatrix:
             return DenseVecMatrix(self._engine().map_op("muls", self._a,
                                                         scalar=other),
                                   self._eng)
+        if isinstance(other, np.ndarray) and other.ndim == 1:
+            # multiply(v: BDV) matrix-vector route (BlockMatrix.scala:265)
+            return self._engine().dgemv(self._a, other)
         if isinstance(other, np.ndarray):
             # multiply(local Breeze matrix) -> DenseVecMatrix
             if self.numCols() != other.shape[0]:
@@ -232,6 +235,13 @@ class BlockMatrix:
         per-tile GEMM-accumulate chain remains (mx_tile_dgemm_acc)."""
         if isinstance(other, (int, float)):
             return self._ew1("muls", other)
+        if isinstance(other, np.ndarray) and other.ndim == 1:
+            # multiply(v: BDV) (BlockMatrix.scala:265-274)
+            if self.numCols() != other.shape[0]:
+                raise ValueError(
+                    f"matrix columns size {self.numCols()} not support "
+                    f"vector length {other.shape[0]}")
+            return self._engine().dgemv(self.toBreeze(), other)
         if isinstance(other, DenseVecMatrix):
             if self.numCols() != other.numRows():
                 raise ValueError(

@@ -518,6 +518,42 @@ __global__ void transpose_kernel(int64_t m, int64_t n,
 }
 
 // ---------------------------------------------------------------------------
+// Matrix-vector multiply (BlockMatrix.multiply(DistributedVector/BDV),
+// BlockMatrix.scala:240-274): y = A x, col-major A. HBM-bound; row-slab
+// blocks x column chunks with fp64 partials, reduced in ascending chunk
+// order (deterministic). Reads of A are fully coalesced per column.
+template <typename T>
+__global__ void gemv_partial_kernel(int64_t m, int64_t n, int64_t lda,
+                                    const T* __restrict__ A,
+                                    const T* __restrict__ x,
+                                    double* __restrict__ partial,
+                                    int nchunks) {
+    int64_t nrb = (m + 255) / 256;
+    int64_t rb = blockIdx.x % nrb;
+    int chunk = (int)(blockIdx.x / nrb);
+    int64_t r = rb * 256 + threadIdx.x;
+    int64_t clen = (n + nchunks - 1) / nchunks;
+    int64_t c0 = chunk * clen;
+    int64_t c1 = c0 + clen < n ? c0 + clen : n;
+    if (r >= m) return;
+    double acc = 0;
+    for (int64_t j = c0; j < c1; j++)
+        acc += (double)A[j * lda + r] * (double)x[j];
+    partial[(int64_t)chunk * m + r] = acc;
+}
+
+template <typename T>
+__global__ void gemv_reduce_kernel(int64_t m, int nchunks,
+                                   const double* __restrict__ partial,
+                                   T* __restrict__ y) {
+    int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (r >= m) return;
+    double acc = 0;
+    for (int c = 0; c < nchunks; c++) acc += partial[(int64_t)c * m + r];
+    y[r] = (T)acc;
+}
+
+// ---------------------------------------------------------------------------
 // C-visible launchers (called from marlin_gpu.cpp).
 extern "C" {
 
@@ -627,6 +663,27 @@ int mxk_transpose(int is_fp32, int64_t m, int64_t n, const void* in,
     else
         hipLaunchKernelGGL(transpose_kernel<double>, grid, block, 0, stream,
                            m, n, (const double*)in, (double*)out);
+    return (int)hipGetLastError() == 0 ? 0 : -2;
+}
+
+int mxk_gemv(int is_fp32, int64_t m, int64_t n, int64_t lda, const void* A,
+             const void* x, double* partial, void* y, hipStream_t stream) {
+    int64_t nrb = (m + 255) / 256;
+    int nchunks = 8;
+    dim3 grid((unsigned)(nrb * nchunks)), block(256);
+    if (is_fp32) {
+        hipLaunchKernelGGL(gemv_partial_kernel<float>, grid, block, 0, stream,
+                           m, n, lda, (const float*)A, (const float*)x,
+                           partial, nchunks);
+        hipLaunchKernelGGL(gemv_reduce_kernel<float>, dim3((unsigned)nrb),
+                           block, 0, stream, m, nchunks, partial, (float*)y);
+    } else {
+        hipLaunchKernelGGL(gemv_partial_kernel<double>, grid, block, 0, stream,
+                           m, n, lda, (const double*)A, (const double*)x,
+                           partial, nchunks);
+        hipLaunchKernelGGL(gemv_reduce_kernel<double>, dim3((unsigned)nrb),
+                           block, 0, stream, m, nchunks, partial, (double*)y);
+    }
     return (int)hipGetLastError() == 0 ? 0 : -2;
 }
 

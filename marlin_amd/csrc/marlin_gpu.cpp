@@ -44,6 +44,8 @@ int mxk_sum(int is_fp32, int64_t n, const void* a, double* partials,
             double* out, hipStream_t stream);
 int mxk_transpose(int is_fp32, int64_t m, int64_t n, const void* in,
                   void* out, hipStream_t stream);
+int mxk_gemv(int is_fp32, int64_t m, int64_t n, int64_t lda, const void* A,
+             const void* x, double* partial, void* y, hipStream_t stream);
 }
 
 #define HIP_OK(x)                                                        \
@@ -657,6 +659,28 @@ int mx_transpose(mx_ctx* c, int is_fp32, int64_t m, int64_t n, const void* A,
     return MX_EHIP;
   HIP_OK(hipMemcpyAsync(C, c->wsB.ptr, m * n * elem, hipMemcpyDeviceToHost,
                         c->s_gemm));
+  HIP_OK(hipStreamSynchronize(c->s_gemm));
+  return MX_OK;
+}
+
+// Matrix-vector multiply (BlockMatrix.scala:240-274): y = A x.
+int mx_dgemv(mx_ctx* c, int64_t m, int64_t n, const double* A,
+             const double* x, double* y) {
+  if (!c || !A || !x || !y || m <= 0 || n <= 0) return MX_EINVAL;
+  HIP_OK(hipSetDevice(c->device));
+  int rc;
+  if ((rc = ensure(c, &c->wsA, m * n * 8))) return rc;
+  if ((rc = ensure(c, &c->wsB, (n + m) * 8))) return rc;
+  if ((rc = ensure(c, &c->wsC, 8 * m * 8))) return rc;  // 8 chunk partials
+  HIP_OK(hipMemcpyAsync(c->wsA.ptr, A, m * n * 8, hipMemcpyHostToDevice,
+                        c->s_gemm));
+  HIP_OK(hipMemcpyAsync(c->wsB.ptr, x, n * 8, hipMemcpyHostToDevice,
+                        c->s_gemm));
+  double* dy = (double*)c->wsB.ptr + n;
+  if ((rc = mxk_gemv(0, m, n, m, c->wsA.ptr, c->wsB.ptr, (double*)c->wsC.ptr,
+                     dy, c->s_gemm)))
+    return MX_EHIP;
+  HIP_OK(hipMemcpyAsync(y, dy, m * 8, hipMemcpyDeviceToHost, c->s_gemm));
   HIP_OK(hipStreamSynchronize(c->s_gemm));
   return MX_OK;
 }

@@ -88,6 +88,7 @@ def _load():
                                   vp, dbl, vp]),
         "mx_sum": (ctypes.c_int, [vp, ctypes.c_int, i64, vp, P(dbl)]),
         "mx_transpose": (ctypes.c_int, [vp, ctypes.c_int, i64, i64, vp, vp]),
+        "mx_dgemv": (ctypes.c_int, [vp, i64, i64, P(dbl), P(dbl), P(dbl)]),
         "mx_stats": (ctypes.c_int, [vp, P(MxStats)]),
     }
     for name, (res, args) in sigs.items():
@@ -271,6 +272,22 @@ class Engine:
                          float(scalar), C.ctypes.data_as(ctypes.c_void_p)),
             "mx_map")
         return C
+
+    def dgemv(self, A, x):
+        """y = A x (BlockMatrix.multiply(DistributedVector) replacement)."""
+        A = np.asfortranarray(A, dtype=np.float64)
+        x = np.ascontiguousarray(x, dtype=np.float64)
+        m, n = A.shape
+        if n != x.shape[0]:
+            raise ValueError(
+                f"matrix columns size {n} not support vector length "
+                f"{x.shape[0]}")
+        y = np.empty(m, dtype=np.float64)
+        _ck(lib().mx_dgemv(self._ctx, m, n, _fbuf(A, np.float64),
+                           x.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+                           y.ctypes.data_as(ctypes.POINTER(ctypes.c_double))),
+            "mx_dgemv")
+        return y
 
     def sum(self, A):
         A = np.asfortranarray(A, dtype=np.float64)

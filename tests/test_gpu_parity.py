@@ -184,3 +184,23 @@ def test_full_size_property_20000(eng):
         eng.free(dA)
         eng.free(dB)
         eng.free(dC)
+
+
+@pytest.mark.parametrize("mn", [(100, 100), (517, 301), (4096, 1000), (1, 7)])
+def test_dgemv_parity(eng, mn):
+    # BlockMatrix.multiply(DistributedVector/BDV) replacement (mx_dgemv)
+    m, n = mn
+    a = gen_matrix(m, n, seed=41)
+    x = gen_matrix(n, 1, seed=42)[:, 0]
+    got = eng.dgemv(a, x)
+    ref = a @ x
+    assert rel_err(got, ref) < 1e-10
+
+
+def test_dgemv_api_routes(eng):
+    a = gen_matrix(64, 48, seed=43)
+    x = gen_matrix(48, 1, seed=44)[:, 0]
+    ref = a @ x
+    assert rel_err(DenseVecMatrix(a, engine=eng).multiply(x), ref) < 1e-10
+    blk = DenseVecMatrix(a, engine=eng).toBlockMatrix(2, 2)
+    assert rel_err(blk.multiply(x), ref) < 1e-10

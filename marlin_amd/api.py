@@ -179,6 +179,16 @@ class DenseVecMatrix:
         from .io import save_matrix_file
         save_matrix_file(self, path)
 
+    def luDecompose(self, mode="auto", base_size=1000):
+        """DenseVecMatrix.luDecompose (DenseVecMatrix.scala:283-464)."""
+        from .linalg import lu_decompose
+        return lu_decompose(self, mode, base_size)
+
+    def inverse(self, mode="auto", base_size=1000):
+        """DenseVecMatrix.inverse (DenseVecMatrix.scala:565-764)."""
+        from .linalg import inverse
+        return inverse(self, mode, base_size)
+
     def toBreeze(self):
         return self._a.copy()
 
@@ -342,6 +352,11 @@ class BlockMatrix:
             save_block_matrix_file(self, path)
         else:
             save_matrix_file(self, path)
+
+    def inverse(self, mode="auto", base_size=1000):
+        """BlockMatrix.inverse (BlockMatrix.scala:527-530):
+        delegates via toDenseVecMatrix."""
+        return self.toDenseVecMatrix().inverse(mode, base_size)
 
     def toDenseVecMatrix(self):
         """BlockMatrix.toDenseVecMatrix (BlockMatrix.scala:575-594)."""

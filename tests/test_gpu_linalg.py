@@ -1,0 +1,62 @@
+# GPU tests for the decomposition tier (blocked LU + inverse on engine
+# GEMMs — DenseVecMatrix.scala:283-464 / 565-764 restatement).
+import numpy as np
+import pytest
+
+from marlin_amd import Engine, DenseVecMatrix
+from oracle import gen_matrix
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def eng():
+    e = Engine(0)
+    yield e
+    e.close()
+
+
+def well_conditioned(n, seed):
+    a = gen_matrix(n, n, seed=seed)
+    return a + n * np.eye(n)
+
+
+def test_inverse_golden_permutation(eng):
+    # DistributedMatrixSuite.scala:340-352: inverse of the flip
+    # permutation is itself (local route, n <= 6000 -> "auto" = local)
+    p3 = np.array([[0., 0, 1], [0, 1, 0], [1, 0, 0]])
+    got = DenseVecMatrix(p3, engine=eng).inverse()
+    np.testing.assert_allclose(got.toBreeze(), p3, atol=1e-12)
+
+
+@pytest.mark.parametrize("n,base", [(300, 96), (517, 128), (64, 16)])
+def test_inverse_dist_route(eng, n, base):
+    a = well_conditioned(n, seed=n)
+    inv = DenseVecMatrix(a, engine=eng).inverse(mode="dist",
+                                                base_size=base).toBreeze()
+    err = np.max(np.abs(a @ inv - np.eye(n)))
+    assert err < 1e-9, err
+
+
+@pytest.mark.parametrize("n,base", [(300, 96), (250, 64)])
+def test_lu_dist_route(eng, n, base):
+    a = well_conditioned(n, seed=1000 + n)
+    dvm = DenseVecMatrix(a, engine=eng)
+    blk, p_array = dvm.luDecompose(mode="dist", base_size=base)
+    lu = blk.toBreeze()
+    L = np.tril(lu, -1) + np.eye(n)
+    U = np.triu(lu)
+    # block pairwise pivoting property: P_blockdiag A == L U
+    PA = a[p_array, :]
+    rel = np.max(np.abs(PA - L @ U)) / np.max(np.abs(a))
+    assert rel < 1e-10, rel
+
+
+def test_lu_local_route(eng):
+    a = well_conditioned(100, seed=77)
+    blk, p_array = DenseVecMatrix(a, engine=eng).luDecompose(mode="breeze")
+    lu = blk.toBreeze()
+    L = np.tril(lu, -1) + np.eye(100)
+    U = np.triu(lu)
+    rel = np.max(np.abs(a[p_array, :] - L @ U)) / np.max(np.abs(a))
+    assert rel < 1e-12, rel

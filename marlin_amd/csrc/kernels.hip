@@ -563,7 +563,10 @@ int mxk_gemm(int is_fp32, int beta_one,
              void* C, int64_t ldc, hipStream_t stream) {
     if (M % 128 || N % 128 || K % 16) return -4;
     int nbm = (int)(M / 128), nbn = (int)(N / 128);
-    int band = nbn < 8 ? nbn : 8;
+    static const char* bandenv = getenv("MARLIN_GEMM_BAND");
+    int bandw = bandenv ? atoi(bandenv) : 8;
+    if (bandw < 1) bandw = 8;
+    int band = nbn < bandw ? nbn : bandw;
     // default: supertiled bands; MARLIN_GEMM_REMAP=bands -> plain bands
     static const char* remap = getenv("MARLIN_GEMM_REMAP");
     if (remap && remap[0] == 'b') band = -band;

@@ -66,18 +66,18 @@ DEVFN void glds16(const void* g, void* lds) {
 // (wave grid 2 x NWAVES/2). BK=32/NWAVES=8 = one 512-thread block per CU
 // (128 KB LDS), half the barrier rate of BK=16/NWAVES=4 (two 256-thread
 // blocks per CU, 64 KB LDS each); both keep 2 waves/SIMD.
-template <typename T, int BETA, int BK, int NWAVES>
-__launch_bounds__(NWAVES * 64, 2)
+template <typename T, int BETA, int BK, int BM, int WM, int WN>
+__launch_bounds__(WM * WN * 64, 2)
 __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
                                  const T* __restrict__ A, int64_t lda,
                                  const T* __restrict__ B, int64_t ldb,
                                  T* __restrict__ C, int64_t ldc,
-                                 int nbm /* grid rows = M/128 */,
+                                 int nbm /* grid rows = M/BM */,
                                  int band /* column-band width in blocks */) {
-    constexpr int BM = 128, BN = 128;
+    constexpr int BN = 128;
+    constexpr int NWAVES = WM * WN;
     constexpr int E = 16 / sizeof(T);       // elems per 16B glds chunk
-    constexpr int WN = NWAVES / 2;          // wave grid 2 x WN
-    constexpr int MT = (BM / 2) / 16;       // 16x16 frags per wave, m
+    constexpr int MT = (BM / WM) / 16;      // 16x16 frags per wave, m
     constexpr int NT = (BN / WN) / 16;      // 16x16 frags per wave, n
     using ACC = typename acc_t<T>::type;
 
@@ -111,7 +111,7 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wid = tid >> 6;
-    const int wm = wid / WN, wn = wid % WN;  // wave grid coords
+    const int wm = wid / WN, wn = wid % WN;  // wave grid coords (WM x WN)
     const int l15 = lane & 15, l4 = lane >> 4;   // MFMA fragment coords
 
     // --- LDS: A as [BK][BM] (k-col major), B as [BN][BK] (n-col major) ---
@@ -122,8 +122,10 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
     const int64_t row0 = (int64_t)bm * BM;    // global row of tile
     const int64_t col0 = (int64_t)bn * BN;    // global col of tile
 
-    constexpr int A_COLS_PER_GLDS = (64 * E) / BM;
-    constexpr int A_GLDS = BK / A_COLS_PER_GLDS / NWAVES;
+    constexpr int A_COLS_PER_GLDS = (64 * E) / BM > 0 ? (64 * E) / BM : 1;
+    constexpr int A_GLDS = (BM * sizeof(T) >= 1024)
+        ? (BK * (int)(BM * sizeof(T) / 1024) / NWAVES)
+        : (BK / A_COLS_PER_GLDS / NWAVES);
     constexpr int B_COLS_PER_GLDS = (64 * E) / BK;
     constexpr int B_GLDS = BN / B_COLS_PER_GLDS / NWAVES;
     // B swizzle parameters: chunk index within a column is XORed with
@@ -144,17 +146,37 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
     //        (BK=16: column stride 128B; BK=32: column stride 256B).
     auto issue_tile = [&](int kt, int buf) {
         const int64_t kbase = (int64_t)kt * BK;
-        #pragma unroll
-        for (int i = 0; i < A_GLDS; i++) {
-            int c = (wid * A_GLDS + i) * A_COLS_PER_GLDS;
-            int lane_col = lane / (BM / E);            // 0 or extra col
-            int lane_row;
-            if constexpr (E == 2)
-                lane_row = (lane ^ (((c + lane_col) & 1) << 3)) * E;
-            else  // f32: 16B chunk (4 elems) XOR 4 by k parity (banks mod 32)
-                lane_row = ((lane % (BM / E)) ^ (((c + lane_col) & 1) << 2)) * E;
-            const T* g = A + (kbase + c + lane_col) * lda + row0 + lane_row;
-            glds16(g, &As[buf * BK * BM + c * BM]);
+        if constexpr (BM * sizeof(T) >= 1024) {
+            // >=1 glds per column: chunk q covers column q/CPC, piece q%CPC
+            constexpr int CPC = BM * sizeof(T) / 1024;
+            constexpr int A_CH = BK * CPC / NWAVES;
+            #pragma unroll
+            for (int i = 0; i < A_CH; i++) {
+                int q = wid * A_CH + i;
+                int c = q / CPC, half = q - (q / CPC) * CPC;
+                int lane_row;
+                if constexpr (E == 2)
+                    lane_row = (half * 64 + (lane ^ ((c & 1) << 3))) * E;
+                else
+                    lane_row = (half * (1024 / sizeof(T) / E) * E)
+                               + ((lane ^ ((c & 1) << 2)) * E);
+                const T* g = A + (kbase + c) * lda + row0 + lane_row;
+                glds16(g, &As[buf * BK * BM + c * BM
+                              + half * (int)(1024 / sizeof(T))]);
+            }
+        } else {
+            #pragma unroll
+            for (int i = 0; i < A_GLDS; i++) {
+                int c = (wid * A_GLDS + i) * A_COLS_PER_GLDS;
+                int lane_col = lane / (BM / E);        // 0 or extra col
+                int lane_row;
+                if constexpr (E == 2)
+                    lane_row = (lane ^ (((c + lane_col) & 1) << 3)) * E;
+                else  // f32: 16B chunk XOR 4 by k parity (banks mod 32)
+                    lane_row = ((lane % (BM / E)) ^ (((c + lane_col) & 1) << 2)) * E;
+                const T* g = A + (kbase + c + lane_col) * lda + row0 + lane_row;
+                glds16(g, &As[buf * BK * BM + c * BM]);
+            }
         }
         #pragma unroll
         for (int i = 0; i < B_GLDS; i++) {
@@ -562,7 +584,7 @@ int mxk_gemm(int is_fp32, int beta_one,
              const void* A, int64_t lda, const void* B, int64_t ldb,
              void* C, int64_t ldc, hipStream_t stream) {
     if (M % 128 || N % 128 || K % 16) return -4;
-    int nbm = (int)(M / 128), nbn = (int)(N / 128);
+    int nbn = (int)(N / 128);
     static const char* bandenv = getenv("MARLIN_GEMM_BAND");
     int bandw = bandenv ? atoi(bandenv) : 8;
     if (bandw < 1) bandw = 8;
@@ -570,22 +592,27 @@ int mxk_gemm(int is_fp32, int beta_one,
     // default: supertiled bands; MARLIN_GEMM_REMAP=bands -> plain bands
     static const char* remap = getenv("MARLIN_GEMM_REMAP");
     if (remap && remap[0] == 'b') band = -band;
-    dim3 grid((unsigned)(nbm * nbn));
-    // config: BK=16 / 2x256-thread blocks per CU is the measured winner
-    // (67.3 vs 58.6 TF at 20000^3); MARLIN_GEMM_CFG=bk32 selects the
-    // 512-thread BK=32 variant (kept for re-evaluation).
+    // configs: default BK=16/BM=128 2x256-thread blocks per CU (measured
+    // winner); MARLIN_GEMM_CFG=bk32 -> 512-thread BK=32; =bm256 -> 256x128
+    // tile / 512-thread (requires M % 256 == 0; halves DMA-per-flop and
+    // barrier rate at 1 block/CU).
     static const char* cfg = getenv("MARLIN_GEMM_CFG");
-    bool bk32 = (K % 32 == 0) && cfg && cfg[0] == 'b' && cfg[2] == '3';
-    #define LAUNCH(TY, BETA, BK, NW)                                       \
-        hipLaunchKernelGGL((gemm_mfma_kernel<TY, BETA, BK, NW>), grid,     \
-            dim3(NW * 64), 0, stream, M, N, K, (const TY*)A, lda,          \
-            (const TY*)B, ldb, (TY*)C, ldc, nbm, band)
+    bool bk32 = (K % 32 == 0) && cfg && cfg[0] == 'b' && cfg[1] == 'k';
+    bool bm256 = (M % 256 == 0) && !is_fp32 && cfg && cfg[0] == 'b' && cfg[1] == 'm';
+    #define LAUNCH(TY, BETA, BK, BMv, WMv, WNv)                            \
+        hipLaunchKernelGGL((gemm_mfma_kernel<TY, BETA, BK, BMv, WMv, WNv>),\
+            dim3((unsigned)((M / BMv) * nbn)), dim3(WMv * WNv * 64), 0,    \
+            stream, M, N, K, (const TY*)A, lda, (const TY*)B, ldb, (TY*)C, \
+            ldc, (int)(M / BMv), band)
     if (is_fp32) {
-        if (beta_one) { if (bk32) LAUNCH(float, 1, 32, 8); else LAUNCH(float, 1, 16, 4); }
-        else          { if (bk32) LAUNCH(float, 0, 32, 8); else LAUNCH(float, 0, 16, 4); }
+        if (beta_one) { if (bk32) LAUNCH(float, 1, 32, 128, 2, 4); else LAUNCH(float, 1, 16, 128, 2, 2); }
+        else          { if (bk32) LAUNCH(float, 0, 32, 128, 2, 4); else LAUNCH(float, 0, 16, 128, 2, 2); }
+    } else if (bm256) {
+        if (beta_one) LAUNCH(double, 1, 16, 256, 4, 2);
+        else          LAUNCH(double, 0, 16, 256, 4, 2);
     } else {
-        if (beta_one) { if (bk32) LAUNCH(double, 1, 32, 8); else LAUNCH(double, 1, 16, 4); }
-        else          { if (bk32) LAUNCH(double, 0, 32, 8); else LAUNCH(double, 0, 16, 4); }
+        if (beta_one) { if (bk32) LAUNCH(double, 1, 32, 128, 2, 4); else LAUNCH(double, 1, 16, 128, 2, 2); }
+        else          { if (bk32) LAUNCH(double, 0, 32, 128, 2, 4); else LAUNCH(double, 0, 16, 128, 2, 2); }
     }
     #undef LAUNCH
     return (int)hipGetLastError() == 0 ? 0 : -2;

@@ -102,7 +102,10 @@ def main():
 
     fp32 = args.dtype == "f32"
     elem = 4 if fp32 else 8
-    eng = Engine(local_rank)
+    # MARLIN_FORCE_DEV0: place every rank on device 0 (multi-rank RCCL
+    # smoke on a single-GPU box; never set by the driver)
+    dev = 0 if os.environ.get("MARLIN_FORCE_DEV0") else local_rank
+    eng = Engine(dev)
 
     # --- distributed setup + device-resident inputs ----------------------
     if world > 1:

@@ -170,6 +170,17 @@ class DenseVecMatrix:
         return DenseVecMatrix(self._engine().map_op(op2, self._a, other._a),
                               self._eng)
 
+    # -- slicing (DenseVecMatrix.sliceByRow/sliceByColumn/getSubMatrix;
+    #    inclusive index ranges, DistributedMatrixSuite.scala:207-224) ----
+    def sliceByRow(self, start, end):
+        return DenseVecMatrix(self._a[start:end + 1, :], self._eng)
+
+    def sliceByColumn(self, start, end):
+        return DenseVecMatrix(self._a[:, start:end + 1], self._eng)
+
+    def getSubMatrix(self, r0, r1, c0, c1):
+        return DenseVecMatrix(self._a[r0:r1 + 1, c0:c1 + 1], self._eng)
+
     def toBlockMatrix(self, blks_by_row, blks_by_col):
         """DenseVecMatrix.toBlockMatrix (DenseVecMatrix.scala:1259-1328)."""
         return _to_block(self._a, blks_by_row, blks_by_col, self._eng)

@@ -148,3 +148,22 @@ def test_gen_matrix_deterministic_colmajor():
     # element (r,c) is a pure function of linear index c*rows+r
     big = gen_matrix(6, 1, seed=1)
     np.testing.assert_array_equal(a.T.reshape(-1)[:3], big[:3, 0])
+
+
+def test_api_slices_golden():
+    # DistributedMatrixSuite.scala:207-224 (host-side slicing; no GPU)
+    from marlin_amd import DenseVecMatrix
+    mat = DenseVecMatrix(M4)
+    np.testing.assert_array_equal(mat.sliceByRow(1, 2).toBreeze(),
+                                  [[2.0, 3, 4, 5], [3, 2, 1, 0]])
+    np.testing.assert_array_equal(mat.sliceByColumn(1, 2).toBreeze(),
+                                  [[1.0, 2], [3, 4], [2, 1], [1, 1]])
+    np.testing.assert_array_equal(mat.getSubMatrix(1, 2, 1, 2).toBreeze(),
+                                  [[3.0, 4], [2, 1]])
+
+
+def test_api_empty_rows_raises():
+    # "empty rows" behaviour (DistributedMatrixSuite.scala:54-63)
+    from marlin_amd import DenseVecMatrix
+    with pytest.raises(RuntimeError):
+        DenseVecMatrix(np.zeros((0, 0)))

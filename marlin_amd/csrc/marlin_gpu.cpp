@@ -537,14 +537,21 @@ static int summa_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
     }
     HIP_OK(hipEventRecord(c->ev[12 + buf], c->s_comm));
 
-    // gemm stream: wait for the panel, accumulate
+    // gemm stream: wait for the panel, accumulate (ev[2/3]+[4/5] pairs
+    // time the gemm and comm legs per buffer; accumulated after sync)
     HIP_OK(hipStreamWaitEvent(c->s_gemm, c->ev[12 + buf], 0));
     int beta = p == 0 ? 0 : 1;
+    HIP_OK(hipEventRecord(c->ev[0], c->s_gemm));
     rc = mxk_gemm(is_fp32, beta, mip, njp, kbp, pa, mip, pb, kbp, dC, mip,
                   c->s_gemm);
     if (rc) return rc == -4 ? MX_EINVAL : MX_EHIP;
     c->st.gemm_launches += 1;
+    HIP_OK(hipEventRecord(c->ev[1], c->s_gemm));
     HIP_OK(hipEventRecord(c->ev[8 + buf], c->s_gemm));
+    HIP_OK(hipEventSynchronize(c->ev[1]));
+    float ms = 0;
+    HIP_OK(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
+    c->st.gemm_ms += ms;
   }
   HIP_OK(hipStreamSynchronize(c->s_gemm));
   HIP_OK(hipStreamSynchronize(c->s_comm));

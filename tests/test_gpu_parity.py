@@ -204,3 +204,47 @@ def test_dgemv_api_routes(eng):
     assert rel_err(DenseVecMatrix(a, engine=eng).multiply(x), ref) < 1e-10
     blk = DenseVecMatrix(a, engine=eng).toBlockMatrix(2, 2)
     assert rel_err(blk.multiply(x), ref) < 1e-10
+
+
+def test_full_size_property_40000_fp32(eng):
+    # BASELINE config 5 size (40000^2 fp32) on one GPU, size-independent
+    # property: C[:,0] == A @ B[:,0] vs a streamed fp64 host recompute,
+    # fp32 bar 1e-4 rel.
+    m = k = n = 40000
+    mp = kp = np_ = 40064   # roundup(40000,128); also mult of 16
+    dA = eng.alloc(mp * kp * 4)
+    dB = eng.alloc(kp * np_ * 4)
+    dC = eng.alloc(mp * np_ * 4)
+    try:
+        eng.fill_random(dA, mp * kp, 0xA11CE, fp32=True)
+        eng.fill_random(dB, kp * np_, 0xB0B, fp32=True)
+        eng.sgemm_device(mp, kp, np_, dA, mp, dB, kp, dC, mp)
+        col = np.empty(mp, dtype=np.float32)
+        eng.download(col, dC, mp * 4)
+        bcol = np.empty(kp, dtype=np.float32)
+        eng.download(bcol, dB, kp * 4)
+        from oracle import gen_uniform_u64
+        acc = np.zeros(m, dtype=np.float64)
+        chunk = 256
+        bcol64 = bcol.astype(np.float64)
+        for c0 in range(0, k, chunk):
+            c1 = min(c0 + chunk, k)
+            z = gen_uniform_u64(0xA11CE, c0 * mp, (c1 - c0) * mp)
+            blockA = ((z >> np.uint64(11)).astype(np.float64) * 2.0 ** -53
+                      ).astype(np.float32).astype(np.float64)
+            blockA = blockA.reshape((c1 - c0, mp)).T[:m]
+            acc += blockA @ bcol64[c0:c1]
+        # pad k-columns beyond k: kp == 40064 > k -> device A cols k..kp
+        # contain random fill that DOES contribute; account for them
+        z = gen_uniform_u64(0xA11CE, k * mp, (kp - k) * mp)
+        blockA = ((z >> np.uint64(11)).astype(np.float64) * 2.0 ** -53
+                  ).astype(np.float32).astype(np.float64)
+        blockA = blockA.reshape((kp - k, mp)).T[:m]
+        acc += blockA @ bcol64[k:kp]
+        rel = np.max(np.abs(col[:m].astype(np.float64) - acc)) / \
+            np.max(np.abs(acc))
+        assert rel < 1e-4, rel
+    finally:
+        eng.free(dA)
+        eng.free(dB)
+        eng.free(dC)

@@ -248,3 +248,26 @@ def test_full_size_property_40000_fp32(eng):
         eng.free(dA)
         eng.free(dB)
         eng.free(dC)
+
+
+def test_dispatch_block_route_and_reblock(eng):
+    # a1: dispatch with a tiny broadcast threshold forces the near-square
+    # block route (DenseVecMatrix.scala:207-213) -> BlockMatrix result
+    m4 = np.array([[0., 1, 2, 3], [2, 3, 4, 5], [3, 2, 1, 0], [1, 1, 1, 1]])
+    c4 = np.array([[11., 10, 9, 8], [23, 24, 25, 26], [7, 11, 15, 19],
+                   [6, 7, 8, 9]])
+    dvm = DenseVecMatrix(m4, engine=eng)
+    res = dvm.multiply(DenseVecMatrix(m4, engine=eng), cores=2,
+                       broadcast_threshold=0)
+    from marlin_amd.api import BlockMatrix as BM
+    assert isinstance(res, BM)
+    assert rel_err(res.toBreeze(), c4) == 0.0
+    # a13: mismatched blocking -> re-block then multiply
+    # (BlockMatrix.scala:187-216 re-slice; re-block equivalence,
+    # DistributedMatrixSuite.scala:411-418)
+    a = gen_matrix(10, 9, seed=51)
+    b = gen_matrix(9, 7, seed=52)
+    blk_a = DenseVecMatrix(a, engine=eng).toBlockMatrix(2, 3)
+    blk_b = DenseVecMatrix(b, engine=eng).toBlockMatrix(2, 2)  # 3 != 2
+    got = blk_a.multiply(blk_b)
+    assert rel_err(got.toBreeze(), a @ b) < 1e-10

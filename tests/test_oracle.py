@@ -167,3 +167,11 @@ def test_api_empty_rows_raises():
     from marlin_amd import DenseVecMatrix
     with pytest.raises(RuntimeError):
         DenseVecMatrix(np.zeros((0, 0)))
+
+
+def test_blockid_hash_eq():
+    # Block.scala:37-49: hash = row*31 + column + seq
+    from marlin_amd import BlockID
+    assert hash(BlockID(2, 3, 4)) == 2 * 31 + 3 + 4
+    assert BlockID(1, 2, 3) == BlockID(1, 2, 3)
+    assert BlockID(1, 2, 3) != BlockID(2, 1, 3)

@@ -223,6 +223,10 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
     const int ntiles = (int)(K / BK);
     issue_tile(0, 0);
 
+    // Two barriers per k-step with the next tile's DMA issued EARLY and
+    // kept in flight across the first barrier via a counted vmcnt (a
+    // single-barrier variant with vmcnt(0)+late issue measured ~1% slower
+    // fp64 / 4% slower fp32 - less DMA lead time).
     for (int kt = 0; kt < ntiles; kt++) {
         const int buf = kt & 1;
         if (kt + 1 < ntiles) {

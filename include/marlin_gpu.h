@@ -134,6 +134,18 @@ int mx_sgemm_device(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
                     const mx_dbuf* dA, int64_t lda,
                     const mx_dbuf* dB, int64_t ldb,
                     mx_dbuf* dC, int64_t ldc);
+/* 2D pitched transfers + zeroing for device-resident matrices (the
+ * RDD.cache() analog) and a beta-capable device GEMM. */
+int mx_upload2d(mx_ctx* ctx, mx_dbuf* dst, int64_t pitch_elems,
+                const void* src, int64_t m, int64_t n, int elem);
+int mx_download2d(mx_ctx* ctx, void* dst, const mx_dbuf* src,
+                  int64_t pitch_elems, int64_t m, int64_t n, int elem);
+int mx_memset(mx_ctx* ctx, mx_dbuf* buf, int64_t bytes);
+int mx_gemm_device_ex(mx_ctx* ctx, int is_fp32, int beta_one, int64_t m,
+                      int64_t k, int64_t n, const mx_dbuf* dA, int64_t lda,
+                      const mx_dbuf* dB, int64_t ldb, mx_dbuf* dC,
+                      int64_t ldc);
+
 /* SUMMA on device-resident local shards (bench hot loop: inputs already
  * in HBM when the timed region starts). */
 int mx_dgemm_summa_device(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,

@@ -55,8 +55,15 @@ class DenseVecMatrix:
     Backed by one ndarray here (single-node host); the engine shards
     on-device. Multiply semantics mirror the reference dispatch."""
 
-    def __init__(self, rows, engine=None):
-        # rows: ndarray (m x n) or dict {row_index: 1-D array}
+    def __init__(self, rows, engine=None, _dev=None):
+        # rows: ndarray (m x n) or dict {row_index: 1-D array}; _dev: a
+        # DeviceMatrix for results materialised lazily (host copy made
+        # only at toBreeze / IO time — the RDD laziness analog)
+        if _dev is not None:
+            self._a = None
+            self._dev = _dev
+            self._eng = engine
+            return
         if isinstance(rows, dict):
             n = 1 + max(rows)
             rows = np.vstack([rows[i] for i in range(n)])
@@ -64,6 +71,7 @@ class DenseVecMatrix:
         if self._a.size == 0:
             raise RuntimeError("empty rows")
         self._eng = engine
+        self._dev = None
 
     def _engine(self):
         if self._eng is None:
@@ -71,10 +79,29 @@ class DenseVecMatrix:
         return self._eng
 
     def numRows(self):
-        return self._a.shape[0]
+        return self._dev.m if self._a is None else self._a.shape[0]
 
     def numCols(self):
-        return self._a.shape[1]
+        return self._dev.n if self._a is None else self._a.shape[1]
+
+    def cache(self):
+        """Keep this matrix resident in HBM (RDD.cache(),
+        DenseVecMatrix.scala:321,505): subsequent multiplies skip the
+        host->device copy."""
+        if self._dev is None:
+            self._dev = self._engine().upload_matrix(self._host())
+        return self
+
+    def unpersist(self):
+        if self._dev is not None and self._a is not None:
+            self._dev.free()
+            self._dev = None
+        return self
+
+    def _host(self):
+        if self._a is None:
+            self._a = self._eng.download_matrix(self._dev)
+        return self._a
 
     def multiply(self, other, cores=None, broadcast_threshold=300):
         """DenseVecMatrix.multiply(other, cores, broadcastThreshold)
@@ -86,19 +113,19 @@ class DenseVecMatrix:
         if isinstance(other, (int, float)):
             # multiply(b: Double) scalar overload (reference test
             # `mat.multiply(2)`, DistributedMatrixSuite.scala:194)
-            return DenseVecMatrix(self._engine().map_op("muls", self._a,
+            return DenseVecMatrix(self._engine().map_op("muls", self._host(),
                                                         scalar=other),
                                   self._eng)
         if isinstance(other, np.ndarray) and other.ndim == 1:
             # multiply(v: BDV) matrix-vector route (BlockMatrix.scala:265)
-            return self._engine().dgemv(self._a, other)
+            return self._engine().dgemv(self._host(), other)
         if isinstance(other, np.ndarray):
             # multiply(local Breeze matrix) -> DenseVecMatrix
             if self.numCols() != other.shape[0]:
                 raise ValueError(
                     f"Dimension mismatch during matrix-matrix multiplication: "
                     f"{self.numCols()} vs {other.shape[0]}")
-            return DenseVecMatrix(self._engine().dgemm(self._a, other),
+            return DenseVecMatrix(self._engine().dgemm(self._host(), other),
                                   self._eng)
         if isinstance(other, BlockMatrix):
             other = other.toDenseVecMatrix()
@@ -108,7 +135,10 @@ class DenseVecMatrix:
                 f"{self.numCols()} vs {other.numRows()}")
         m, k, n = self.numRows(), self.numCols(), other.numCols()
         bsize = broadcast_threshold * 1024 * 1024 // 8
-        c = self._engine().dgemm(self._a, other._a)
+        if self._dev is not None and other._dev is not None:
+            cdev = self._engine().gemm_dd(self._dev, other._dev)
+            return DenseVecMatrix(None, self._eng, _dev=cdev)
+        c = self._engine().dgemm(self._host(), other._host())
         if k * n <= bsize or m * k <= bsize:
             return DenseVecMatrix(c, self._eng)
         if (0.8 < (m * n) / (k * k) < 1.2) and (0.8 < m / k < 1.2):
@@ -127,19 +157,19 @@ class DenseVecMatrix:
         return self._ew2("sub", "subs", other)
 
     def subtractBy(self, b):
-        return DenseVecMatrix(self._engine().map_op("rsubs", self._a,
+        return DenseVecMatrix(self._engine().map_op("rsubs", self._host(),
                                                     scalar=b), self._eng)
 
     def multiply_scalar(self, b):
-        return DenseVecMatrix(self._engine().map_op("muls", self._a,
+        return DenseVecMatrix(self._engine().map_op("muls", self._host(),
                                                     scalar=b), self._eng)
 
     def divide(self, b):
-        return DenseVecMatrix(self._engine().map_op("divs", self._a,
+        return DenseVecMatrix(self._engine().map_op("divs", self._host(),
                                                     scalar=b), self._eng)
 
     def divideBy(self, b):
-        return DenseVecMatrix(self._engine().map_op("rdivs", self._a,
+        return DenseVecMatrix(self._engine().map_op("rdivs", self._host(),
                                                     scalar=b), self._eng)
 
     def dotProduct(self, other):
@@ -148,18 +178,18 @@ class DenseVecMatrix:
         if (self.numRows() != other.numRows()
                 or self.numCols() != other.numCols()):
             raise ValueError("matrix dimension mismatch")
-        return DenseVecMatrix(self._engine().map_op("emul", self._a,
-                                                    other._a), self._eng)
+        return DenseVecMatrix(self._engine().map_op("emul", self._host(),
+                                                    other._host()), self._eng)
 
     def sum(self):
-        return self._engine().sum(self._a)
+        return self._engine().sum(self._host())
 
     def transpose(self):
-        return DenseVecMatrix(self._engine().transpose(self._a), self._eng)
+        return DenseVecMatrix(self._engine().transpose(self._host()), self._eng)
 
     def _ew2(self, op2, op1, other):
         if isinstance(other, (int, float)):
-            return DenseVecMatrix(self._engine().map_op(op1, self._a,
+            return DenseVecMatrix(self._engine().map_op(op1, self._host(),
                                                         scalar=other),
                                   self._eng)
         if isinstance(other, BlockMatrix):
@@ -167,23 +197,24 @@ class DenseVecMatrix:
         if (self.numRows() != other.numRows()
                 or self.numCols() != other.numCols()):
             raise ValueError("matrix dimension mismatch")
-        return DenseVecMatrix(self._engine().map_op(op2, self._a, other._a),
+        return DenseVecMatrix(self._engine().map_op(op2, self._host(),
+                                                    other._host()),
                               self._eng)
 
     # -- slicing (DenseVecMatrix.sliceByRow/sliceByColumn/getSubMatrix;
     #    inclusive index ranges, DistributedMatrixSuite.scala:207-224) ----
     def sliceByRow(self, start, end):
-        return DenseVecMatrix(self._a[start:end + 1, :], self._eng)
+        return DenseVecMatrix(self._host()[start:end + 1, :], self._eng)
 
     def sliceByColumn(self, start, end):
-        return DenseVecMatrix(self._a[:, start:end + 1], self._eng)
+        return DenseVecMatrix(self._host()[:, start:end + 1], self._eng)
 
     def getSubMatrix(self, r0, r1, c0, c1):
-        return DenseVecMatrix(self._a[r0:r1 + 1, c0:c1 + 1], self._eng)
+        return DenseVecMatrix(self._host()[r0:r1 + 1, c0:c1 + 1], self._eng)
 
     def toBlockMatrix(self, blks_by_row, blks_by_col):
         """DenseVecMatrix.toBlockMatrix (DenseVecMatrix.scala:1259-1328)."""
-        return _to_block(self._a, blks_by_row, blks_by_col, self._eng)
+        return _to_block(self._host(), blks_by_row, blks_by_col, self._eng)
 
     def saveToFileSystem(self, path):
         """DenseVecMatrix.saveToFileSystem (DenseVecMatrix.scala:1042)."""
@@ -201,7 +232,7 @@ class DenseVecMatrix:
         return inverse(self, mode, base_size)
 
     def toBreeze(self):
-        return self._a.copy()
+        return self._host().copy()
 
 
 def _to_block(a, r, c, engine):
@@ -270,7 +301,7 @@ class BlockMatrix:
                     f"{self.numCols()} vs {other.numRows()}")
             # BlockMatrix x DenseVecMatrix (BlockMatrix.scala:305-335)
             return DenseVecMatrix(
-                self._engine().dgemm(self.toBreeze(), other._a), self._eng)
+                self._engine().dgemm(self.toBreeze(), other._host()), self._eng)
         if self.numCols() != other.numRows():
             raise ValueError(
                 f"Dimension mismatch during matrix-matrix multiplication: "
@@ -314,7 +345,7 @@ class BlockMatrix:
 
     def dotProduct(self, other):
         if isinstance(other, DenseVecMatrix):
-            other = _to_block(other._a, self._nbr, self._nbc, self._eng)
+            other = _to_block(other._host(), self._nbr, self._nbc, self._eng)
         if (self.numRows() != other.numRows()
                 or self.numCols() != other.numCols()):
             raise ValueError("matrix dimension mismatch")
@@ -343,7 +374,7 @@ class BlockMatrix:
         if isinstance(other, (int, float)):
             return self._ew1(op1, other)
         if isinstance(other, DenseVecMatrix):
-            other = _to_block(other._a, self._nbr, self._nbc, self._eng)
+            other = _to_block(other._host(), self._nbr, self._nbc, self._eng)
         if (self.numRows() != other.numRows()
                 or self.numCols() != other.numCols()):
             raise ValueError("matrix dimension mismatch")

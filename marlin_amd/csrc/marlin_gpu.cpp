@@ -299,6 +299,45 @@ int mx_sgemm_device(mx_ctx* c, int64_t m, int64_t k, int64_t n,
                      ldc);
 }
 
+// 2D pitched transfers for device-resident matrices (the RDD.cache()
+// analog: DenseVecMatrix.cache(), DenseVecMatrix.scala:321,505 usage).
+int mx_upload2d(mx_ctx* c, mx_dbuf* dst, int64_t pitch_elems,
+                const void* src, int64_t m, int64_t n, int elem) {
+  if (!c || !dst || !src || m <= 0 || n <= 0) return MX_EINVAL;
+  if (pitch_elems * (n - 1) + m > dst->bytes / elem) return MX_EINVAL;
+  HIP_OK(hipSetDevice(c->device));
+  HIP_OK(hipMemcpy2D(dst->ptr, (size_t)(pitch_elems * elem), src,
+                     (size_t)(m * elem), (size_t)(m * elem), (size_t)n,
+                     hipMemcpyHostToDevice));
+  return MX_OK;
+}
+int mx_download2d(mx_ctx* c, void* dst, const mx_dbuf* src,
+                  int64_t pitch_elems, int64_t m, int64_t n, int elem) {
+  if (!c || !dst || !src || m <= 0 || n <= 0) return MX_EINVAL;
+  HIP_OK(hipSetDevice(c->device));
+  HIP_OK(hipMemcpy2D(dst, (size_t)(m * elem), src->ptr,
+                     (size_t)(pitch_elems * elem), (size_t)(m * elem),
+                     (size_t)n, hipMemcpyDeviceToHost));
+  return MX_OK;
+}
+int mx_memset(mx_ctx* c, mx_dbuf* b, int64_t bytes) {
+  if (!c || !b || bytes > b->bytes) return MX_EINVAL;
+  HIP_OK(hipSetDevice(c->device));
+  HIP_OK(hipMemset(b->ptr, 0, (size_t)bytes));
+  return MX_OK;
+}
+// beta-capable device-resident GEMM (C += A*B when beta_one)
+int mx_gemm_device_ex(mx_ctx* c, int is_fp32, int beta_one, int64_t m,
+                      int64_t k, int64_t n, const mx_dbuf* dA, int64_t lda,
+                      const mx_dbuf* dB, int64_t ldb, mx_dbuf* dC,
+                      int64_t ldc) {
+  if (!c || !dA || !dB || !dC) return MX_EINVAL;
+  c->st = {};
+  c->st.flops = 2.0 * m * k * n;
+  return gemm_device(c, is_fp32, beta_one, m, k, n, dA->ptr, lda, dB->ptr,
+                     ldb, dC->ptr, ldc);
+}
+
 // ---------------------------------------------------------------------------
 // host-buffer whole-multiply entries: pad -> H2D -> kernel -> D2H.
 // elem = 8 (fp64) or 4 (fp32).

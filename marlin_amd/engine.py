@@ -89,6 +89,14 @@ def _load():
         "mx_sum": (ctypes.c_int, [vp, ctypes.c_int, i64, vp, P(dbl)]),
         "mx_transpose": (ctypes.c_int, [vp, ctypes.c_int, i64, i64, vp, vp]),
         "mx_dgemv": (ctypes.c_int, [vp, i64, i64, P(dbl), P(dbl), P(dbl)]),
+        "mx_upload2d": (ctypes.c_int, [vp, vp, i64, vp, i64, i64,
+                                       ctypes.c_int]),
+        "mx_download2d": (ctypes.c_int, [vp, vp, vp, i64, i64, i64,
+                                         ctypes.c_int]),
+        "mx_memset": (ctypes.c_int, [vp, vp, i64]),
+        "mx_gemm_device_ex": (ctypes.c_int, [vp, ctypes.c_int, ctypes.c_int,
+                                             i64, i64, i64, vp, i64, vp, i64,
+                                             vp, i64]),
         "mx_stats": (ctypes.c_int, [vp, P(MxStats)]),
     }
     for name, (res, args) in sigs.items():
@@ -143,6 +151,28 @@ def _fbuf(a, dtype):
     assert a.dtype == dtype and a.flags.f_contiguous
     return a.ctypes.data_as(ctypes.POINTER(
         ctypes.c_double if dtype == np.float64 else ctypes.c_float))
+
+
+class DeviceMatrix:
+    """A matrix resident in HBM (the RDD.cache() analog): logical m x n,
+    col-major with GEMM-ready padded pitch (rows padded to 128, pad
+    region zeroed). Owned by an Engine; freed via .free() or engine
+    close."""
+
+    __slots__ = ("eng", "buf", "m", "n", "pitch", "fp32")
+
+    def __init__(self, eng, buf, m, n, pitch, fp32=False):
+        self.eng, self.buf = eng, buf
+        self.m, self.n, self.pitch, self.fp32 = m, n, pitch, fp32
+
+    @property
+    def elem(self):
+        return 4 if self.fp32 else 8
+
+    def free(self):
+        if self.buf is not None:
+            self.eng.free(self.buf)
+            self.buf = None
 
 
 class Engine:
@@ -311,6 +341,57 @@ class Engine:
         st = MxStats()
         _ck(lib().mx_stats(self._ctx, ctypes.byref(st)))
         return {f: getattr(st, f) for f, _ in MxStats._fields_}
+
+    # -- device-resident matrices (RDD.cache() analog) ----------------------
+    def upload_matrix(self, a, fp32=False):
+        dt = np.float32 if fp32 else np.float64
+        a = np.asfortranarray(a, dtype=dt)
+        m, n = a.shape
+        pitch = (m + 127) // 128 * 128
+        # pad columns to 128 so the matrix can stand as the GEMM rhs
+        # (n-dim padded to the 128 tile) as well as lhs (k padded to 16);
+        # pad region zeroed -> padded-dim GEMM stays exact
+        ncap = (n + 127) // 128 * 128
+        elem = 4 if fp32 else 8
+        buf = self.alloc(pitch * ncap * elem)
+        _ck(lib().mx_memset(self._ctx, buf, pitch * ncap * elem))
+        _ck(lib().mx_upload2d(self._ctx, buf, pitch,
+                              a.ctypes.data_as(ctypes.c_void_p), m, n, elem),
+            "mx_upload2d")
+        return DeviceMatrix(self, buf, m, n, pitch, fp32)
+
+    def download_matrix(self, dm):
+        dt = np.float32 if dm.fp32 else np.float64
+        out = np.empty((dm.m, dm.n), dtype=dt, order="F")
+        _ck(lib().mx_download2d(self._ctx, out.ctypes.data_as(ctypes.c_void_p),
+                                dm.buf, dm.pitch, dm.m, dm.n, dm.elem),
+            "mx_download2d")
+        return out
+
+    def gemm_dd(self, A, B, C=None, accumulate=False):
+        """Device-resident C (+)= A @ B on cached matrices. Pads are
+        zero, so padded-dim GEMM is exact; result C is a DeviceMatrix
+        with logical (A.m, B.n)."""
+        assert A.fp32 == B.fp32
+        if A.n != B.m:
+            raise ValueError(
+                f"Dimension mismatch during matrix-matrix multiplication: "
+                f"{A.n} vs {B.m}")
+        kp = (A.n + 15) // 16 * 16
+        np_ = (B.n + 127) // 128 * 128
+        if C is None:
+            assert not accumulate
+            cbuf = self.alloc(A.pitch * np_ * A.elem)
+            C = DeviceMatrix(self, cbuf, A.m, B.n, A.pitch, A.fp32)
+        assert C.pitch == A.pitch and C.m == A.m and C.n == B.n
+        # B must expose kp padded rows: its pitch is >= kp iff B.m pads ok
+        assert B.pitch >= kp
+        _ck(lib().mx_gemm_device_ex(self._ctx, 1 if A.fp32 else 0,
+                                    1 if accumulate else 0,
+                                    A.pitch, kp, np_, A.buf, A.pitch,
+                                    B.buf, B.pitch, C.buf, C.pitch),
+            "mx_gemm_device_ex")
+        return C
 
     # -- device-resident bench path ------------------------------------------
     def alloc(self, nbytes):

@@ -271,3 +271,40 @@ def test_dispatch_block_route_and_reblock(eng):
     blk_b = DenseVecMatrix(b, engine=eng).toBlockMatrix(2, 2)  # 3 != 2
     got = blk_a.multiply(blk_b)
     assert rel_err(got.toBreeze(), a @ b) < 1e-10
+
+
+def test_cached_device_resident_multiply(eng):
+    # RDD.cache() analog: cached operands multiply device-resident (no
+    # per-op PCIe round trip), result materialises lazily at toBreeze()
+    a = gen_matrix(300, 500, seed=61)
+    b = gen_matrix(500, 260, seed=62)
+    dvm_a = DenseVecMatrix(a, engine=eng).cache()
+    dvm_b = DenseVecMatrix(b, engine=eng).cache()
+    res = dvm_a.multiply(dvm_b, broadcast_threshold=10 ** 9)
+    ref = a @ b
+    assert rel_err(res.toBreeze(), ref) < 1e-10
+    # chained: (A@B) result is itself device-resident; re-cache-free reuse
+    res2 = dvm_a.multiply(dvm_b, broadcast_threshold=10 ** 9)
+    assert rel_err(res2.toBreeze(), ref) < 1e-10
+    # cached -> uncached mixed route still correct
+    res3 = dvm_a.multiply(DenseVecMatrix(b, engine=eng))
+    assert rel_err(res3.toBreeze(), ref) < 1e-10
+    dvm_a.unpersist()
+    dvm_b.unpersist()
+
+
+def test_gemm_dd_accumulate(eng):
+    # device-resident accumulate chain (C += A@B)
+    a1 = gen_matrix(100, 90, seed=63)
+    b1 = gen_matrix(90, 110, seed=64)
+    a2 = gen_matrix(100, 70, seed=65)
+    b2 = gen_matrix(70, 110, seed=66)
+    A1, B1 = eng.upload_matrix(a1), eng.upload_matrix(b1)
+    A2, B2 = eng.upload_matrix(a2), eng.upload_matrix(b2)
+    C = eng.gemm_dd(A1, B1)
+    C = eng.gemm_dd(A2, B2, C, accumulate=True)
+    got = eng.download_matrix(C)
+    ref = a1 @ b1 + a2 @ b2
+    assert rel_err(got, ref) < 1e-10
+    for d in (A1, B1, A2, B2, C):
+        d.free()

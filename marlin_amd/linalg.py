@@ -6,8 +6,11 @@
 # Structure mirrors the reference exactly: the diagonal base block is
 # factored on the HOST (the reference factors it on the Spark driver with
 # Breeze LU/inv — a deliberate host step, not a fallback), and every
-# panel scale and trailing-matrix update is a dense GEMM on the engine
-# (the reference's executor-side SubMatrix.multiply chain).
+# panel scale and trailing-matrix update is a dense GEMM on the engine.
+# All blocks stay DEVICE-RESIDENT (DeviceMatrix) across the sweeps —
+# the RDD.cache() analog — so only the base-sized diagonal factors cross
+# PCIe; sign flips are folded into the small host factors so no device
+# elementwise pass is needed.
 #
 # luDecompose: block pairwise pivoting (pivoting INSIDE each diagonal
 # block only — the reference's scheme). Returns (blocks, p_array) with
@@ -21,16 +24,18 @@ import math
 import numpy as np
 
 
-def _ceil_block(total, parts):
-    return int(math.ceil(total / parts))
-
-
 def _split(n, base):
     nb = int(math.ceil(n / base))
-    bl = _ceil_block(n, nb)
+    bl = int(math.ceil(n / nb))
     offs = [i * bl for i in range(nb)]
     lens = [min(bl, n - o) for o in offs]
     return nb, offs, lens
+
+
+def _upload_blocks(eng, a, nb, offs, lens):
+    return {(i, j): eng.upload_matrix(a[offs[i]:offs[i] + lens[i],
+                                        offs[j]:offs[j] + lens[j]])
+            for i in range(nb) for j in range(nb)}
 
 
 def lu_decompose(dvm, mode="auto", base_size=1000):
@@ -57,46 +62,61 @@ def lu_decompose(dvm, mode="auto", base_size=1000):
 
     eng = dvm._engine()
     nb, offs, lens = _split(n, base_size)
-    blk = {(i, j): a[offs[i]:offs[i] + lens[i],
-                     offs[j]:offs[j] + lens[j]].copy()
-           for i in range(nb) for j in range(nb)}
+    blk = _upload_blocks(eng, a, nb, offs, lens)
+    host = {}                       # finished blocks, host-side
     p_array = np.arange(n)
 
     for i in range(nb):
-        lu, piv = scipy.linalg.lu_factor(blk[(i, i)])
+        diag = eng.download_matrix(blk[(i, i)])
+        lu, piv = scipy.linalg.lu_factor(diag)
         perm = np.arange(lens[i])
         for r, p in enumerate(piv):
             perm[r], perm[p] = perm[p], perm[r]
         p_array[offs[i]:offs[i] + lens[i]] = offs[i] + perm
-        blk[(i, i)] = lu                      # packed L\U of the diagonal
+        host[(i, i)] = lu                     # packed L\U of the diagonal
         if i == nb - 1:
             break
         L = np.tril(lu, -1) + np.eye(lens[i])
         U = np.triu(lu)
         P = np.zeros((lens[i], lens[i]))
         P[np.arange(lens[i]), perm] = 1.0     # row g of P·X is X[perm[g]]
-        Minv_l = np.linalg.solve(L, P)        # L^-1 P   (host, base-sized)
-        Uinv = np.linalg.inv(U)               # U^-1     (host, base-sized)
-        # panel scales + trailing update — engine GEMMs
+        d_Minv = eng.upload_matrix(np.linalg.solve(L, P))     # L^-1 P
+        d_negMinv = eng.upload_matrix(-np.linalg.solve(L, P))
+        d_Uinv = eng.upload_matrix(np.linalg.inv(U))          # U^-1
+        # panel scales + trailing update — device-resident GEMMs
+        negA2 = {}
         for j in range(i + 1, nb):
-            blk[(i, j)] = eng.dgemm(Minv_l, blk[(i, j)])   # U12 panel
+            negA2[j] = eng.gemm_dd(d_negMinv, blk[(i, j)])    # -L^-1 P A12
+            new = eng.gemm_dd(d_Minv, blk[(i, j)])            # U12 panel
+            blk[(i, j)].free()
+            blk[(i, j)] = new
         for r in range(i + 1, nb):
-            blk[(r, i)] = eng.dgemm(blk[(r, i)], Uinv)     # L21 panel
+            new = eng.gemm_dd(blk[(r, i)], d_Uinv)            # L21 panel
+            blk[(r, i)].free()
+            blk[(r, i)] = new
         for r in range(i + 1, nb):
-            neg_l = -blk[(r, i)]
             for j in range(i + 1, nb):
-                # A22 -= L21 U12  (reference: A4 - A3 (A11 \\ A2))
-                blk[(r, j)] = eng.tile_dgemm_acc(neg_l, blk[(i, j)],
-                                                 blk[(r, j)])
-    # sub-diagonal permutation fix-up (DenseVecMatrix.scala:444-462):
-    # L21 block rows permuted by their OWN block row's perm
+                # A22 += L21 · (-U12)  (reference: A4 - A3 (A11 \\ A2))
+                eng.gemm_dd(blk[(r, i)], negA2[j], blk[(r, j)],
+                            accumulate=True)
+        for d in negA2.values():
+            d.free()
+        d_Minv.free(); d_negMinv.free(); d_Uinv.free()
+
+    # materialise + sub-diagonal permutation fix-up
+    # (DenseVecMatrix.scala:444-462): L21 block rows permuted by their
+    # OWN block row's perm
+    for (i, j), d in blk.items():
+        if (i, j) not in host:
+            host[(i, j)] = eng.download_matrix(d)
+        d.free()
     for r in range(1, nb):
         perm = p_array[offs[r]:offs[r] + lens[r]] - offs[r]
         P = np.zeros((lens[r], lens[r]))
         P[np.arange(lens[r]), perm] = 1.0
         for c in range(r):
-            blk[(r, c)] = P @ blk[(r, c)]
-    return (BlockMatrix(blk, n, n, engine=dvm._eng), p_array)
+            host[(r, c)] = P @ host[(r, c)]
+    return (BlockMatrix(host, n, n, engine=dvm._eng), p_array)
 
 
 def inverse(dvm, mode="auto", base_size=1000):
@@ -117,51 +137,60 @@ def inverse(dvm, mode="auto", base_size=1000):
 
     eng = dvm._engine()
     nb, offs, lens = _split(n, base_size)
-    blk = {(i, j): a[offs[i]:offs[i] + lens[i],
-                     offs[j]:offs[j] + lens[j]].copy()
-           for i in range(nb) for j in range(nb)}
+    blk = _upload_blocks(eng, a, nb, offs, lens)
     S1, S2, S3 = {}, {}, {}
 
     # forward sweep (DenseVecMatrix.scala:604-675)
     for i in range(nb - 1):
-        inv = np.linalg.inv(blk[(i, i)])      # host base block (driver inv)
+        inv = np.linalg.inv(eng.download_matrix(blk[(i, i)]))
         S1[i] = inv
+        d_inv = eng.upload_matrix(inv)
+        d_neg = eng.upload_matrix(-inv)
         for j in range(i + 1, nb):
-            S2[(i, j)] = -eng.dgemm(inv, blk[(i, j)])      # -A11^-1 A12
+            S2[(i, j)] = eng.gemm_dd(d_neg, blk[(i, j)])      # -A11^-1 A12
         for r in range(i + 1, nb):
-            S3[(r, i)] = -eng.dgemm(blk[(r, i)], inv)      # -A21 A11^-1
-        for j in range(i + 1, nb):
-            t = eng.dgemm(inv, blk[(i, j)])                # A11^-1 A12
-            for r in range(i + 1, nb):
-                neg = -blk[(r, i)]
-                blk[(r, j)] = eng.tile_dgemm_acc(neg, t, blk[(r, j)])
-    # last trailing block
-    T = {(nb - 1, nb - 1): np.linalg.inv(blk[(nb - 1, nb - 1)])}
+            S3[(r, i)] = eng.gemm_dd(blk[(r, i)], d_neg)      # -A21 A11^-1
+        for r in range(i + 1, nb):
+            for j in range(i + 1, nb):
+                # A22 += A21 · (-A11^-1 A12)
+                eng.gemm_dd(blk[(r, i)], S2[(i, j)], blk[(r, j)],
+                            accumulate=True)
+        d_inv.free(); d_neg.free()
+    last = eng.download_matrix(blk[(nb - 1, nb - 1)])
+    T = {(nb - 1, nb - 1): eng.upload_matrix(np.linalg.inv(last))}
 
-    # backward sweep (DenseVecMatrix.scala:677-764)
+    # backward sweep (DenseVecMatrix.scala:677-764) — device-resident
     for i in range(nb - 2, -1, -1):
         col = {}
         for r in range(i + 1, nb):
             acc = None
             for c in range(i + 1, nb):
-                acc = eng.tile_dgemm_acc(T[(r, c)], S3[(c, i)], acc)
-            col[r] = acc                                    # inv[r, i]
+                if acc is None:
+                    acc = eng.gemm_dd(T[(r, c)], S3[(c, i)])
+                else:
+                    eng.gemm_dd(T[(r, c)], S3[(c, i)], acc, accumulate=True)
+            col[r] = acc                                      # inv[r, i]
         row = {}
         for c in range(i + 1, nb):
             acc = None
             for j in range(i + 1, nb):
-                acc = eng.tile_dgemm_acc(S2[(i, j)], T[(j, c)], acc)
-            row[c] = acc                                    # inv[i, c]
-        corner = S1[i].copy()
+                if acc is None:
+                    acc = eng.gemm_dd(S2[(i, j)], T[(j, c)])
+                else:
+                    eng.gemm_dd(S2[(i, j)], T[(j, c)], acc, accumulate=True)
+            row[c] = acc                                      # inv[i, c]
+        corner = eng.upload_matrix(S1[i])
         for j in range(i + 1, nb):
-            corner = eng.tile_dgemm_acc(S2[(i, j)], col[j], corner)
-        newT = {(i, i): corner}
+            eng.gemm_dd(S2[(i, j)], col[j], corner, accumulate=True)
+        T[(i, i)] = corner
         for r in range(i + 1, nb):
-            newT[(r, i)] = col[r]
+            T[(r, i)] = col[r]
         for c in range(i + 1, nb):
-            newT[(i, c)] = row[c]
-        for r in range(i + 1, nb):
-            for c in range(i + 1, nb):
-                newT[(r, c)] = T[(r, c)]
-        T = newT
-    return BlockMatrix(T, n, n, engine=dvm._eng)
+            T[(i, c)] = row[c]
+
+    out = {k: eng.download_matrix(d) for k, d in T.items()}
+    for d in T.values():
+        d.free()
+    for d in list(S2.values()) + list(S3.values()) + list(blk.values()):
+        d.free()
+    return BlockMatrix(out, n, n, engine=dvm._eng)

@@ -22,6 +22,7 @@
 #include <cstdlib>
 #include <vector>
 #include <algorithm>
+#include <utility>
 
 #include "../../include/marlin_gpu.h"
 
@@ -517,7 +518,10 @@ static int summa_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
   const int64_t mip = round_up(mi, 128), njp = round_up(nj, 128);
   // local shard pitches ARE the padded sizes (bench fills them that way;
   // host entry packs them that way)
-  const int64_t kb_max = 4096;
+  // panel width: overlap granularity of the comm/MFMA pipeline
+  // (MARLIN_SUMMA_KB for round-2 8-GPU tuning; default 4096)
+  static const char* kbenv = getenv("MARLIN_SUMMA_KB");
+  const int64_t kb_max = kbenv && atoll(kbenv) > 0 ? atoll(kbenv) : 4096;
   auto panels = plan_panels(k, c->pr, c->pc, kb_max);
 
   c->st = {};
@@ -535,6 +539,18 @@ static int summa_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
   // ev[8+b]: gemm done reading panel buffer b; ev[12+b]: panel b ready
   HIP_OK(hipEventRecord(c->ev[8], c->s_gemm));
   HIP_OK(hipEventRecord(c->ev[9], c->s_gemm));
+
+  // deferred per-panel timing: events recorded in-loop, synchronised
+  // only AFTER both streams drain (an in-loop sync would serialise the
+  // comm/GEMM double-buffer pipeline)
+  std::vector<hipEvent_t> tev;
+  auto mkev = [&]() {
+    hipEvent_t e;
+    (void)hipEventCreate(&e);
+    tev.push_back(e);
+    return e;
+  };
+  std::vector<std::pair<hipEvent_t, hipEvent_t>> gemm_tv, comm_tv;
 
   for (size_t p = 0; p < panels.size(); p++) {
     const panel_t& pan = panels[p];
@@ -567,33 +583,45 @@ static int summa_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
                               nj, hipMemcpyDeviceToDevice, c->s_comm));
     }
     if (c->nranks > 1) {
+      hipEvent_t c0 = mkev(), c1 = mkev();
+      HIP_OK(hipEventRecord(c0, c->s_comm));
       RCCL_OK(ncclGroupStart());
       RCCL_OK(ncclBroadcast(pa, pa, (size_t)(mip * kbp), nty, pan.rootA,
                             c->rowc, c->s_comm));
       RCCL_OK(ncclBroadcast(pb, pb, (size_t)(kbp * njp), nty, pan.rootB,
                             c->colc, c->s_comm));
       RCCL_OK(ncclGroupEnd());
+      HIP_OK(hipEventRecord(c1, c->s_comm));
+      comm_tv.push_back({c0, c1});
     }
     HIP_OK(hipEventRecord(c->ev[12 + buf], c->s_comm));
 
-    // gemm stream: wait for the panel, accumulate (ev[2/3]+[4/5] pairs
-    // time the gemm and comm legs per buffer; accumulated after sync)
+    // gemm stream: wait for the panel, accumulate
     HIP_OK(hipStreamWaitEvent(c->s_gemm, c->ev[12 + buf], 0));
     int beta = p == 0 ? 0 : 1;
-    HIP_OK(hipEventRecord(c->ev[0], c->s_gemm));
+    hipEvent_t g0 = mkev(), g1 = mkev();
+    HIP_OK(hipEventRecord(g0, c->s_gemm));
     rc = mxk_gemm(is_fp32, beta, mip, njp, kbp, pa, mip, pb, kbp, dC, mip,
                   c->s_gemm);
     if (rc) return rc == -4 ? MX_EINVAL : MX_EHIP;
     c->st.gemm_launches += 1;
-    HIP_OK(hipEventRecord(c->ev[1], c->s_gemm));
+    HIP_OK(hipEventRecord(g1, c->s_gemm));
+    gemm_tv.push_back({g0, g1});
     HIP_OK(hipEventRecord(c->ev[8 + buf], c->s_gemm));
-    HIP_OK(hipEventSynchronize(c->ev[1]));
-    float ms = 0;
-    HIP_OK(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
-    c->st.gemm_ms += ms;
   }
   HIP_OK(hipStreamSynchronize(c->s_gemm));
   HIP_OK(hipStreamSynchronize(c->s_comm));
+  for (auto& pr2 : gemm_tv) {
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, pr2.first, pr2.second);
+    c->st.gemm_ms += ms;
+  }
+  for (auto& pr2 : comm_tv) {
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, pr2.first, pr2.second);
+    c->st.comm_ms += ms;
+  }
+  for (hipEvent_t e : tev) (void)hipEventDestroy(e);
   return MX_OK;
 }
 

@@ -308,3 +308,15 @@ def test_gemm_dd_accumulate(eng):
     assert rel_err(got, ref) < 1e-10
     for d in (A1, B1, A2, B2, C):
         d.free()
+
+
+def test_random_shape_sweep(eng):
+    # broad parity insurance: pseudo-random (m, k, n) shapes, fixed seed
+    rng = np.random.RandomState(0xC0FFEE)
+    for t in range(10):
+        m, k, n = (int(rng.randint(1, 1500)) for _ in range(3))
+        a = gen_matrix(m, k, seed=7000 + t)
+        b = gen_matrix(k, n, seed=8000 + t)
+        got = eng.dgemm(a, b)
+        ref = a @ b
+        assert rel_err(got, ref) < 1e-10, (m, k, n)

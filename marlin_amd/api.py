@@ -294,6 +294,17 @@ class BlockMatrix:
                     f"matrix columns size {self.numCols()} not support "
                     f"vector length {other.shape[0]}")
             return self._engine().dgemv(self.toBreeze(), other)
+        if isinstance(other, np.ndarray):
+            # multiply(B: BDM) broadcast small-B route
+            # (BlockMatrix.scala:280-303): one engine GEMM, result kept
+            # in this matrix's block grid
+            if self.numCols() != other.shape[0]:
+                raise ValueError(
+                    f"Dimension mismatch during matrix-matrix "
+                    f"multiplication: {self.numCols()} vs {other.shape[0]}")
+            c = self._engine().dgemm(self.toBreeze(), other)
+            return _to_block(c, self.numBlksByRow(), self.numBlksByCol(),
+                             self._eng)
         if isinstance(other, DenseVecMatrix):
             if self.numCols() != other.numRows():
                 raise ValueError(

@@ -175,3 +175,26 @@ def test_blockid_hash_eq():
     assert hash(BlockID(2, 3, 4)) == 2 * 31 + 3 + 4
     assert BlockID(1, 2, 3) == BlockID(1, 2, 3)
     assert BlockID(1, 2, 3) != BlockID(2, 1, 3)
+
+
+def test_repeat_by_row_and_column():
+    # DistributedMatrixSuite.scala:355-374 ("repeat by row and column")
+    from marlin_amd import DenseVecMatrix, repeat_by_row, repeat_by_column
+    mat = DenseVecMatrix(M4)
+    np.testing.assert_array_equal(repeat_by_row(mat, 2).toBreeze(),
+                                  np.tile(M4, (1, 2)))
+    np.testing.assert_array_equal(repeat_by_column(mat, 2).toBreeze(),
+                                  np.tile(M4, (2, 1)))
+    blk = mat.toBlockMatrix(2, 2)
+    np.testing.assert_array_equal(repeat_by_row(blk, 2).toBreeze(),
+                                  np.tile(M4, (1, 2)))
+    np.testing.assert_array_equal(repeat_by_column(blk, 3).toBreeze(),
+                                  np.tile(M4, (3, 1)))
+    with pytest.raises(ValueError):
+        repeat_by_row(mat, 0)
+
+
+def test_random_den_vec_matrix_matches_engine_spec():
+    from marlin_amd import random_den_vec_matrix
+    got = random_den_vec_matrix(5, 3, seed=0xA11CE).toBreeze()
+    np.testing.assert_array_equal(got, gen_matrix(5, 3, seed=0xA11CE))

@@ -136,8 +136,10 @@ def main():
         eng.fill_random(dA, mip * kaj, 0xA11CE + rank, fp32)
         eng.fill_random(dB, kbi_p * nj, 0xB0B + rank, fp32)
 
+        summa = eng.sgemm_summa_device if fp32 else eng.dgemm_summa_device
+
         def step():
-            eng.dgemm_summa_device(m, k, n, dA, dB, dC)
+            summa(m, k, n, dA, dB, dC)
     else:
         mp, kp, np_ = roundup(m, 128), roundup(k, 16), roundup(n, 128)
         dA = eng.alloc(mp * kp * elem)

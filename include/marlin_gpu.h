@@ -151,6 +151,9 @@ int mx_gemm_device_ex(mx_ctx* ctx, int is_fp32, int beta_one, int64_t m,
 int mx_dgemm_summa_device(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
                           const mx_dbuf* dA_local, const mx_dbuf* dB_local,
                           mx_dbuf* dC_local);
+int mx_sgemm_summa_device(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
+                          const mx_dbuf* dA_local, const mx_dbuf* dB_local,
+                          mx_dbuf* dC_local);
 
 /* ---- elementwise / reduction / transpose (BlockMatrix epilogue ops,
  * BlockMatrix.scala:344-523; DenseVecMatrix.scala scalar ops) ---------- */

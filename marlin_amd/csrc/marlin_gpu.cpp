@@ -630,6 +630,11 @@ int mx_dgemm_summa_device(mx_ctx* c, int64_t m, int64_t k, int64_t n,
   if (!c || !dA || !dB || !dC) return MX_EINVAL;
   return summa_device(c, 0, m, k, n, dA->ptr, dB->ptr, dC->ptr);
 }
+int mx_sgemm_summa_device(mx_ctx* c, int64_t m, int64_t k, int64_t n,
+                          const mx_dbuf* dA, const mx_dbuf* dB, mx_dbuf* dC) {
+  if (!c || !dA || !dB || !dC) return MX_EINVAL;
+  return summa_device(c, 1, m, k, n, dA->ptr, dB->ptr, dC->ptr);
+}
 
 // host-buffer SUMMA entries: pack local shards padded, H2D, run, D2H
 static int summa_host(mx_ctx* c, int is_fp32, int64_t m, int64_t k, int64_t n,

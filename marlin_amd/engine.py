@@ -84,6 +84,7 @@ def _load():
         "mx_sgemm_device": (ctypes.c_int, [vp, i64, i64, i64, vp, i64, vp, i64,
                                            vp, i64]),
         "mx_dgemm_summa_device": (ctypes.c_int, [vp, i64, i64, i64, vp, vp, vp]),
+        "mx_sgemm_summa_device": (ctypes.c_int, [vp, i64, i64, i64, vp, vp, vp]),
         "mx_map": (ctypes.c_int, [vp, ctypes.c_int, ctypes.c_int, i64, vp,
                                   vp, dbl, vp]),
         "mx_sum": (ctypes.c_int, [vp, ctypes.c_int, i64, vp, P(dbl)]),
@@ -417,6 +418,10 @@ class Engine:
     def dgemm_summa_device(self, m, k, n, dA, dB, dC):
         _ck(lib().mx_dgemm_summa_device(self._ctx, m, k, n, dA, dB, dC),
             "mx_dgemm_summa_device")
+
+    def sgemm_summa_device(self, m, k, n, dA, dB, dC):
+        _ck(lib().mx_sgemm_summa_device(self._ctx, m, k, n, dA, dB, dC),
+            "mx_sgemm_summa_device")
 
     def download(self, host_arr, dbuf, nbytes):
         _ck(lib().mx_download(self._ctx,

@@ -178,6 +178,17 @@ def main():
         tf = flops / dt / 1e12
         ms_per_step = dt * 1000.0 / args.steps
         peak = FP32_PEAK_TF if fp32 else FP64_PEAK_TF
+        # confirm the peak from the box's own CU count x max clock
+        # (SURVEY 8d): fp64 MFMA = 32 FLOP/clk/SIMD x 4 SIMD/CU;
+        # fp32 = 2x that rate
+        try:
+            cus, clk = eng.device_info()
+            flop_per_clk = (64 if fp32 else 32) * 4
+            measured_peak = cus * flop_per_clk * clk * 1e3 / 1e12
+            if 0.5 * peak < measured_peak < 1.5 * peak:
+                peak = round(measured_peak, 2)
+        except Exception:
+            pass
         # dominant-kernel roofline: HIP-event time of the MFMA GEMM kernel
         # launches (stats from the engine's gemm-stream events)
         if launches and gemm_ms_acc > 0:

@@ -63,6 +63,9 @@ int mx_comm_id(char unique_id[MX_UNIQUE_ID_BYTES]);  /* rank 0 calls this */
 int mx_comm_init(mx_ctx* ctx, int rank, int nranks,
                  const char unique_id[MX_UNIQUE_ID_BYTES]);
 
+/* Device facts (CU count, max clock kHz) for on-box peak computation. */
+int mx_device_info(mx_ctx* ctx, int* cus, int* clock_khz);
+
 /* Grid geometry of this rank after mx_comm_init (pr,pc,row,col). */
 int mx_grid(mx_ctx* ctx, int* pr, int* pc, int* prow, int* pcol);
 

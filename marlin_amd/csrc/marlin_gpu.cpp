@@ -179,6 +179,17 @@ int mx_comm_id(char unique_id[MX_UNIQUE_ID_BYTES]) {
 }
 
 static void grid_shape(int nranks, int* pr, int* pc) {
+  // MARLIN_GRID=RxC overrides (e.g. 2x4 for the 8-GPU grid sweep)
+  const char* g = getenv("MARLIN_GRID");
+  if (g) {
+    int r = 0, cc = 0;
+    if (sscanf(g, "%dx%d", &r, &cc) == 2 && r > 0 && cc > 0 &&
+        r * cc == nranks) {
+      *pr = r;
+      *pc = cc;
+      return;
+    }
+  }
   switch (nranks) {
     case 8: *pr = 4; *pc = 2; break;
     case 4: *pr = 2; *pc = 2; break;
@@ -207,6 +218,17 @@ int mx_comm_init(mx_ctx* c, int rank, int nranks,
     c->colc = c->world;
   }
   c->have_comm = true;
+  return MX_OK;
+}
+
+// Device facts for peak computation (confirm the fp64 MFMA
+// peak from CU count x clock on the actual box, not just the 78.6 spec).
+int mx_device_info(mx_ctx* c, int* cus, int* clock_khz) {
+  if (!c) return MX_EINVAL;
+  hipDeviceProp_t prop;
+  HIP_OK(hipGetDeviceProperties(&prop, c->device));
+  if (cus) *cus = prop.multiProcessorCount;
+  if (clock_khz) *clock_khz = prop.clockRate;
   return MX_OK;
 }
 

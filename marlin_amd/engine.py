@@ -59,6 +59,8 @@ def _load():
         "mx_comm_init": (ctypes.c_int, [vp, ctypes.c_int, ctypes.c_int,
                                         ctypes.c_char_p]),
         "mx_grid": (ctypes.c_int, [vp] + [P(ctypes.c_int)] * 4),
+        "mx_device_info": (ctypes.c_int, [vp, P(ctypes.c_int),
+                                          P(ctypes.c_int)]),
         "mx_dgemm": (ctypes.c_int, [vp, i64, i64, i64, P(dbl), P(dbl), P(dbl)]),
         "mx_sgemm": (ctypes.c_int, [vp, i64, i64, i64, P(flt), P(flt), P(flt)]),
         "mx_sgemm_epilogue": (ctypes.c_int, [vp, i64, i64, i64, P(flt), P(flt),
@@ -202,6 +204,13 @@ class Engine:
         assert len(uid_bytes) == UNIQUE_ID_BYTES
         _ck(lib().mx_comm_init(self._ctx, rank, nranks, uid_bytes),
             "mx_comm_init")
+
+    def device_info(self):
+        """(CU count, max clock kHz) of the bound device."""
+        cus, clk = ctypes.c_int(), ctypes.c_int()
+        _ck(lib().mx_device_info(self._ctx, ctypes.byref(cus),
+                                 ctypes.byref(clk)))
+        return cus.value, clk.value
 
     def grid(self):
         vals = [ctypes.c_int() for _ in range(4)]

@@ -320,3 +320,32 @@ def test_random_shape_sweep(eng):
         got = eng.dgemm(a, b)
         ref = a @ b
         assert rel_err(got, ref) < 1e-10, (m, k, n)
+
+
+def test_special_values_propagate(eng):
+    # IEEE special values must propagate exactly as an FMA chain does
+    # (v_mfma f64 is a k-ordered FMA chain; C/D never flush)
+    a = gen_matrix(64, 64, seed=71)
+    b = gen_matrix(64, 64, seed=72)
+    a[3, 5] = np.inf
+    a[10, 11] = -np.inf
+    a[20, 2] = np.nan
+    a[33, 40] = 5e-320          # subnormal
+    got = eng.dgemm(a, b)
+    ref = a @ b
+    # NaN/Inf pattern identical
+    np.testing.assert_array_equal(np.isnan(got), np.isnan(ref))
+    np.testing.assert_array_equal(np.isinf(got), np.isinf(ref))
+    finite = np.isfinite(ref)
+    assert np.max(np.abs(got[finite] - ref[finite])) / \
+        np.max(np.abs(ref[finite])) < 1e-10
+
+
+def test_long_k_accumulation(eng):
+    # K = 100000: 6250 K-loop iterations; fp64 error growth ~sqrt(K)*eps
+    m, k, n = 64, 100000, 64
+    a = gen_matrix(m, k, seed=73)
+    b = gen_matrix(k, n, seed=74)
+    got = eng.dgemm(a, b)
+    ref = a @ b
+    assert rel_err(got, ref) < 1e-12

@@ -212,6 +212,26 @@ class DenseVecMatrix:
     def getSubMatrix(self, r0, r1, c0, c1):
         return DenseVecMatrix(self._host()[r0:r1 + 1, c0:c1 + 1], self._eng)
 
+    def elementsCount(self):
+        """DistributedMatrix.elementsCount: row count for DenseVecMatrix."""
+        return self.numRows()
+
+    def cBind(self, other):
+        """DistributedMatrix.cBind: column bind."""
+        if isinstance(other, BlockMatrix):
+            other = other.toDenseVecMatrix()
+        if self.numRows() != other.numRows():
+            raise ValueError("matrix dimension mismatch")
+        return DenseVecMatrix(np.hstack([self._host(), other._host()]),
+                              self._eng)
+
+    def print(self, max_rows=10, max_cols=10):
+        a = self._host()
+        print(a[:max_rows, :max_cols])
+
+    def printAll(self):
+        print(self._host())
+
     def toBlockMatrix(self, blks_by_row, blks_by_col):
         """DenseVecMatrix.toBlockMatrix (DenseVecMatrix.scala:1259-1328)."""
         return _to_block(self._host(), blks_by_row, blks_by_col, self._eng)
@@ -410,6 +430,19 @@ class BlockMatrix:
         """BlockMatrix.inverse (BlockMatrix.scala:527-530):
         delegates via toDenseVecMatrix."""
         return self.toDenseVecMatrix().inverse(mode, base_size)
+
+    def elementsCount(self):
+        """DistributedMatrix.elementsCount: sub-block count."""
+        return len(self._blocks)
+
+    def cBind(self, other):
+        return self.toDenseVecMatrix().cBind(other)
+
+    def print(self, max_rows=10, max_cols=10):
+        print(self.toBreeze()[:max_rows, :max_cols])
+
+    def printAll(self):
+        print(self.toBreeze())
 
     def toDenseVecMatrix(self):
         """BlockMatrix.toDenseVecMatrix (BlockMatrix.scala:575-594)."""

@@ -198,3 +198,22 @@ def test_random_den_vec_matrix_matches_engine_spec():
     from marlin_amd import random_den_vec_matrix
     got = random_den_vec_matrix(5, 3, seed=0xA11CE).toBreeze()
     np.testing.assert_array_equal(got, gen_matrix(5, 3, seed=0xA11CE))
+
+
+def test_trait_surface_complete():
+    # every public method of the reference's DistributedMatrix trait
+    # (DistributedMatrix.scala:9-76) exists on both mirrors
+    from marlin_amd import DenseVecMatrix, BlockMatrix
+    trait = ["numRows", "numCols", "toBreeze", "add", "subtract",
+             "subtractBy", "multiply", "divide", "divideBy",
+             "elementsCount", "sum", "dotProduct", "transpose", "inverse",
+             "cBind", "saveToFileSystem", "print", "printAll"]
+    dvm = DenseVecMatrix(M4)
+    blk = dvm.toBlockMatrix(2, 2)
+    for name in trait:
+        assert hasattr(dvm, name), f"DenseVecMatrix missing {name}"
+        assert hasattr(blk, name), f"BlockMatrix missing {name}"
+    assert dvm.elementsCount() == 4
+    assert blk.elementsCount() == 4
+    both = dvm.cBind(DenseVecMatrix(M4))
+    np.testing.assert_array_equal(both.toBreeze(), np.hstack([M4, M4]))

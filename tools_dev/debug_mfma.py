@@ -24,17 +24,9 @@ C = eng.dgemm(I, B)
 print("I@B == B:", np.allclose(C, B))
 print("I@B == B.T:", np.allclose(C, B.T))
 if not np.allclose(C, B):
-    # locate permutation: where does row 0 of B land?
     idx = np.argmax(np.abs(C - B) > 1e-12)
     r, c = np.unravel_index(idx, C.shape)
     print("first mismatch at", r, c, "got", C[r, c], "want", B[r, c])
-    # check a few candidate permutations of rows
-    l = np.arange(n)
-    for name, perm in [
-        ("rowswap16: r -> (r%16)*? ", None),
-    ]:
-        pass
-    # dump 18x6 corner
     np.set_printoptions(precision=3, suppress=True, linewidth=200)
     print("C corner:\n", C[:18, :6])
     print("B corner:\n", B[:18, :6])

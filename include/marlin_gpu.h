@@ -144,6 +144,8 @@ int mx_upload2d(mx_ctx* ctx, mx_dbuf* dst, int64_t pitch_elems,
 int mx_download2d(mx_ctx* ctx, void* dst, const mx_dbuf* src,
                   int64_t pitch_elems, int64_t m, int64_t n, int elem);
 int mx_memset(mx_ctx* ctx, mx_dbuf* buf, int64_t bytes);
+int mx_transpose_device(mx_ctx* ctx, int is_fp32, int64_t m, int64_t n,
+                        const mx_dbuf* in, mx_dbuf* out);
 int mx_gemm_device_ex(mx_ctx* ctx, int is_fp32, int beta_one, int64_t m,
                       int64_t k, int64_t n, const mx_dbuf* dA, int64_t lda,
                       const mx_dbuf* dB, int64_t ldb, mx_dbuf* dC,

@@ -251,6 +251,11 @@ class DenseVecMatrix:
         from .linalg import inverse
         return inverse(self, mode, base_size)
 
+    def choleskyDecompose(self, mode="auto", base_size=1000):
+        """DenseVecMatrix.choleskyDecompose (DenseVecMatrix.scala:475-566)."""
+        from .linalg import cholesky_decompose
+        return cholesky_decompose(self, mode, base_size)
+
     def toBreeze(self):
         return self._host().copy()
 

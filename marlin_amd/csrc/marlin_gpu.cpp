@@ -349,6 +349,18 @@ int mx_memset(mx_ctx* c, mx_dbuf* b, int64_t bytes) {
   HIP_OK(hipMemset(b->ptr, 0, (size_t)bytes));
   return MX_OK;
 }
+// device-resident transpose (operates on the full padded image; pads
+// are zero so the transposed pads stay zero)
+int mx_transpose_device(mx_ctx* c, int is_fp32, int64_t m, int64_t n,
+                        const mx_dbuf* in, mx_dbuf* out) {
+  if (!c || !in || !out) return MX_EINVAL;
+  HIP_OK(hipSetDevice(c->device));
+  if (mxk_transpose(is_fp32, m, n, in->ptr, out->ptr, c->s_gemm))
+    return MX_EHIP;
+  HIP_OK(hipStreamSynchronize(c->s_gemm));
+  return MX_OK;
+}
+
 // beta-capable device-resident GEMM (C += A*B when beta_one)
 int mx_gemm_device_ex(mx_ctx* c, int is_fp32, int beta_one, int64_t m,
                       int64_t k, int64_t n, const mx_dbuf* dA, int64_t lda,

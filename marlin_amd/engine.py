@@ -100,6 +100,8 @@ def _load():
         "mx_gemm_device_ex": (ctypes.c_int, [vp, ctypes.c_int, ctypes.c_int,
                                              i64, i64, i64, vp, i64, vp, i64,
                                              vp, i64]),
+        "mx_transpose_device": (ctypes.c_int, [vp, ctypes.c_int, i64, i64,
+                                               vp, vp]),
         "mx_stats": (ctypes.c_int, [vp, P(MxStats)]),
     }
     for name, (res, args) in sigs.items():
@@ -377,6 +379,16 @@ class Engine:
                                 dm.buf, dm.pitch, dm.m, dm.n, dm.elem),
             "mx_download2d")
         return out
+
+    def transpose_dd(self, dm):
+        """Device-resident transpose: returns a GEMM-ready DeviceMatrix
+        (the padded image transposes wholesale; zero pads stay zero)."""
+        ncap = ((dm.n + 127) // 128 * 128)
+        out = self.alloc(ncap * dm.pitch * dm.elem)
+        _ck(lib().mx_transpose_device(self._ctx, 1 if dm.fp32 else 0,
+                                      dm.pitch, ncap, dm.buf, out),
+            "mx_transpose_device")
+        return DeviceMatrix(self, out, dm.n, dm.m, ncap, dm.fp32)
 
     def gemm_dd(self, A, B, C=None, accumulate=False):
         """Device-resident C (+)= A @ B on cached matrices. Pads are

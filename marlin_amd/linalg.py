@@ -194,3 +194,69 @@ def inverse(dvm, mode="auto", base_size=1000):
     for d in list(S2.values()) + list(S3.values()) + list(blk.values()):
         d.free()
     return BlockMatrix(out, n, n, engine=dvm._eng)
+
+
+def cholesky_decompose(dvm, mode="auto", base_size=1000):
+    """DenseVecMatrix.choleskyDecompose (DenseVecMatrix.scala:475-566):
+    A = L L^T, lower-triangular L. Right-looking blocked: host chol of
+    the (symmetrized, as the reference does) diagonal block, panel scale
+    L21 = A21 L11^-T and trailing A22 -= L21 L21^T as device-resident
+    engine GEMMs; signs folded into the host factor."""
+    from .api import BlockMatrix
+
+    a = dvm.toBreeze()
+    n = a.shape[0]
+    if a.shape[0] != a.shape[1]:
+        raise ValueError(
+            f"LU decompose only support square matrix: {n} v.s {n}")
+    if mode == "auto":
+        mode = "dist" if n > 6000 else "local"
+    if mode in ("local", "breeze"):
+        L = np.linalg.cholesky(np.tril(a) + np.tril(a, -1).T)
+        return BlockMatrix({(0, 0): L}, n, n, engine=dvm._eng)
+
+    eng = dvm._engine()
+    nb, offs, lens = _split(n, base_size)
+    blk = {(i, j): eng.upload_matrix(a[offs[i]:offs[i] + lens[i],
+                                       offs[j]:offs[j] + lens[j]])
+           for i in range(nb) for j in range(nb) if i >= j}
+    host = {}
+    for i in range(nb):
+        diag = eng.download_matrix(blk[(i, i)])
+        diag = np.tril(diag) + np.tril(diag, -1).T   # symmetrize (ref does)
+        L11 = np.linalg.cholesky(diag)
+        host[(i, i)] = L11
+        if i == nb - 1:
+            break
+        linvT = np.linalg.inv(L11).T                 # L11^-T (host, base)
+        d_pos = eng.upload_matrix(linvT)
+        d_neg = eng.upload_matrix(-linvT)
+        negT = {}
+        for r in range(i + 1, nb):
+            old = blk[(r, i)]
+            l21 = eng.gemm_dd(old, d_pos)            # L21 panel
+            negp = eng.gemm_dd(old, d_neg)           # -L21
+            negT[r] = eng.transpose_dd(negp)         # -L21^T
+            negp.free()
+            old.free()
+            blk[(r, i)] = l21
+        for c in range(i + 1, nb):
+            for r in range(c, nb):
+                # A22 += L21[r] @ (-L21[c]^T)
+                eng.gemm_dd(blk[(r, i)], negT[c], blk[(r, c)],
+                            accumulate=True)
+        for d in negT.values():
+            d.free()
+        d_pos.free()
+        d_neg.free()
+
+    for (i, j), d in blk.items():
+        if (i, j) not in host:
+            host[(i, j)] = eng.download_matrix(d)
+        d.free()
+    # upper-triangle blocks are zero in the L result
+    for i in range(nb):
+        for j in range(i + 1, nb):
+            host[(i, j)] = np.zeros((lens[i], lens[j]))
+    # keep strictly-lower of diagonal factor only (chol returns lower)
+    return BlockMatrix(host, n, n, engine=dvm._eng)

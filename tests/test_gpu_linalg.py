@@ -60,3 +60,37 @@ def test_lu_local_route(eng):
     U = np.triu(lu)
     rel = np.max(np.abs(a[p_array, :] - L @ U)) / np.max(np.abs(a))
     assert rel < 1e-12, rel
+
+
+@pytest.mark.parametrize("n,base", [(300, 96), (250, 64)])
+def test_cholesky_dist_route(eng, n, base):
+    # SPD input: L L^T must reproduce the symmetrized matrix
+    r = gen_matrix(n, n, seed=2000 + n)
+    a = (r + r.T) / 2 + n * np.eye(n)
+    L = DenseVecMatrix(a, engine=eng).choleskyDecompose(
+        mode="dist", base_size=base).toBreeze()
+    assert np.allclose(np.triu(L, 1), 0)
+    rel = np.max(np.abs(L @ L.T - a)) / np.max(np.abs(a))
+    assert rel < 1e-12, rel
+
+
+def test_cholesky_local_route(eng):
+    r = gen_matrix(60, 60, seed=3000)
+    a = (r + r.T) / 2 + 60 * np.eye(60)
+    L = DenseVecMatrix(a, engine=eng).choleskyDecompose().toBreeze()
+    rel = np.max(np.abs(L @ L.T - a)) / np.max(np.abs(a))
+    assert rel < 1e-12, rel
+
+
+def test_transpose_dd(eng):
+    a = gen_matrix(130, 70, seed=4000)
+    A = eng.upload_matrix(a)
+    At = eng.transpose_dd(A)
+    got = eng.download_matrix(At)
+    np.testing.assert_array_equal(got, a.T)
+    # transposed handle is GEMM-ready: A^T @ A
+    prod = eng.gemm_dd(At, A)
+    np.testing.assert_allclose(eng.download_matrix(prod), a.T @ a,
+                               rtol=1e-12, atol=1e-12)
+    for d in (A, At, prod):
+        d.free()

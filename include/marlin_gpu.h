@@ -182,6 +182,10 @@ int mx_transpose(mx_ctx* ctx, int is_fp32, int64_t m, int64_t n,
 int mx_dgemv(mx_ctx* ctx, int64_t m, int64_t n, const double* A,
              const double* x, double* y);
 
+/* Device-resident gemv on a cached matrix (mx_dbuf + pitch). */
+int mx_dgemv_device(mx_ctx* ctx, int64_t m, int64_t n, const mx_dbuf* dA,
+                    int64_t lda, const double* x, double* y);
+
 /* ---- timing / stats (MTUtils.evaluate + RMMcompare.scala:47-51 analog) -- */
 typedef struct {
   double h2d_ms;          /* host->device copies of the last call           */

@@ -251,6 +251,46 @@ class DenseVecMatrix:
         from .linalg import inverse
         return inverse(self, mode, base_size)
 
+    def lr(self, step_size, iters):
+        """DenseVecMatrix.lr (DenseVecMatrix.scala:1005-1035): SGD
+        logistic-regression gradient sums. Row format (label, features);
+        the first feature element is overwritten with the intercept 1,
+        exactly as the reference does. The per-iteration reduce of
+        per-row gradients equals X'^T (sigmoid(X' w) - labels) — two
+        gemv passes on the cached device matrix per iteration."""
+        a = self._host()
+        labels = a[:, 0].copy()
+        X = a.copy()
+        X[:, 0] = 1.0                      # intercept
+        m, f = X.shape
+        eng = self._engine()
+        dX = eng.upload_matrix(X)
+        dXT = eng.transpose_dd(dX)
+        w = np.zeros(f)
+        for i in range(1, iters + 1):
+            margin = -eng.dgemv_dd(dX, w)
+            gmul = 1.0 / (1.0 + np.exp(margin)) - labels
+            delta = eng.dgemv_dd(dXT, gmul)
+            w = w - delta * (step_size / m / np.sqrt(i))
+        dX.free()
+        dXT.free()
+        return w
+
+    def computeGramianMatrix(self):
+        """DenseVecMatrix.computeGramianMatrix (DenseVecMatrix.scala:1464):
+        the local Gram matrix A^T A (device-resident compute)."""
+        n = self.numCols()
+        if n > 65535:
+            raise ValueError(f"Argument with more than 65535 cols: {n}")
+        eng = self._engine()
+        dA = eng.upload_matrix(self._host())
+        dAT = eng.transpose_dd(dA)
+        G = eng.gemm_dd(dAT, dA)
+        out = eng.download_matrix(G)
+        for d in (dA, dAT, G):
+            d.free()
+        return out
+
     def choleskyDecompose(self, mode="auto", base_size=1000):
         """DenseVecMatrix.choleskyDecompose (DenseVecMatrix.scala:475-566)."""
         from .linalg import cholesky_decompose

@@ -798,6 +798,25 @@ int mx_dgemv(mx_ctx* c, int64_t m, int64_t n, const double* A,
   return MX_OK;
 }
 
+// Device-resident gemv on a cached matrix (x, y host vectors; the
+// matrix never crosses PCIe). lda = the DeviceMatrix pitch.
+int mx_dgemv_device(mx_ctx* c, int64_t m, int64_t n, const mx_dbuf* dA,
+                    int64_t lda, const double* x, double* y) {
+  if (!c || !dA || !x || !y || m <= 0 || n <= 0) return MX_EINVAL;
+  HIP_OK(hipSetDevice(c->device));
+  int rc;
+  if ((rc = ensure(c, &c->wsB, (n + m + 8 * m) * 8))) return rc;
+  double* dx = (double*)c->wsB.ptr;
+  double* dy = dx + n;
+  double* parts = dy + m;
+  HIP_OK(hipMemcpyAsync(dx, x, n * 8, hipMemcpyHostToDevice, c->s_gemm));
+  if ((rc = mxk_gemv(0, m, n, lda, dA->ptr, dx, parts, dy, c->s_gemm)))
+    return MX_EHIP;
+  HIP_OK(hipMemcpyAsync(y, dy, m * 8, hipMemcpyDeviceToHost, c->s_gemm));
+  HIP_OK(hipStreamSynchronize(c->s_gemm));
+  return MX_OK;
+}
+
 int mx_stats(mx_ctx* c, mx_stats_t* out) {
   if (!c || !out) return MX_EINVAL;
   *out = c->st;

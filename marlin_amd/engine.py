@@ -102,6 +102,8 @@ def _load():
                                              vp, i64]),
         "mx_transpose_device": (ctypes.c_int, [vp, ctypes.c_int, i64, i64,
                                                vp, vp]),
+        "mx_dgemv_device": (ctypes.c_int, [vp, i64, i64, vp, i64, P(dbl),
+                                           P(dbl)]),
         "mx_stats": (ctypes.c_int, [vp, P(MxStats)]),
     }
     for name, (res, args) in sigs.items():
@@ -379,6 +381,25 @@ class Engine:
                                 dm.buf, dm.pitch, dm.m, dm.n, dm.elem),
             "mx_download2d")
         return out
+
+    def dgemv_dd(self, dm, x):
+        """y = A x with A device-resident (pads are zero -> exact)."""
+        x = np.ascontiguousarray(x, dtype=np.float64)
+        if dm.n != x.shape[0]:
+            raise ValueError("dimension mismatch")
+        # run over padded rows/cols; x padded with zeros to the pitch cols
+        ncap = ((dm.n + 127) // 128 * 128)
+        xp = np.zeros(ncap, dtype=np.float64)
+        xp[:dm.n] = x
+        y = np.empty(dm.pitch, dtype=np.float64)
+        _ck(lib().mx_dgemv_device(self._ctx, dm.pitch, ncap, dm.buf,
+                                  dm.pitch,
+                                  xp.ctypes.data_as(ctypes.POINTER(
+                                      ctypes.c_double)),
+                                  y.ctypes.data_as(ctypes.POINTER(
+                                      ctypes.c_double))),
+            "mx_dgemv_device")
+        return y[:dm.m].copy()
 
     def transpose_dd(self, dm):
         """Device-resident transpose: returns a GEMM-ready DeviceMatrix

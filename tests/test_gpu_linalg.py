@@ -94,3 +94,39 @@ def test_transpose_dd(eng):
                                rtol=1e-12, atol=1e-12)
     for d in (A, At, prod):
         d.free()
+
+
+def test_gramian(eng):
+    a = gen_matrix(500, 120, seed=5000)
+    g = DenseVecMatrix(a, engine=eng).computeGramianMatrix()
+    np.testing.assert_allclose(g, a.T @ a, rtol=1e-12, atol=1e-12)
+
+
+def test_dgemv_dd(eng):
+    a = gen_matrix(300, 200, seed=5100)
+    x = gen_matrix(200, 1, seed=5101)[:, 0]
+    dA = eng.upload_matrix(a)
+    np.testing.assert_allclose(eng.dgemv_dd(dA, x), a @ x,
+                               rtol=1e-12, atol=1e-12)
+    dA.free()
+
+
+def test_lr_matches_reference_sgd(eng):
+    # reproduce the reference's full-batch gradient loop on the host and
+    # compare weights after a few iterations
+    rows = gen_matrix(200, 6, seed=5200)
+    labels = (rows[:, 1] > 0.5).astype(np.float64)
+    data = rows.copy()
+    data[:, 0] = labels
+    got = DenseVecMatrix(data, engine=eng).lr(0.5, 5)
+    # host restatement (DenseVecMatrix.scala:1005-1035)
+    X = data.copy()
+    X[:, 0] = 1.0
+    w = np.zeros(6)
+    m = X.shape[0]
+    for i in range(1, 6):
+        margin = -(X @ w)
+        gmul = 1.0 / (1.0 + np.exp(margin)) - labels
+        delta = X.T @ gmul
+        w = w - delta * (0.5 / m / np.sqrt(i))
+    np.testing.assert_allclose(got, w, rtol=1e-10, atol=1e-12)

@@ -351,6 +351,20 @@ class Engine:
             "mx_transpose")
         return C
 
+    def sgemm_summa(self, m, k, n, A_local, B_local):
+        """fp32 distributed multiply on this rank's shards."""
+        A_local = np.asfortranarray(A_local, dtype=np.float32)
+        B_local = np.asfortranarray(B_local, dtype=np.float32)
+        pr, pc, prow, pcol = self.grid()
+        mi = slab_len(m, pr, prow)
+        nj = slab_len(n, pc, pcol)
+        C = np.empty((mi, nj), dtype=np.float32, order="F")
+        _ck(lib().mx_sgemm_summa(self._ctx, m, k, n,
+                                 _fbuf(A_local, np.float32),
+                                 _fbuf(B_local, np.float32),
+                                 _fbuf(C, np.float32)), "mx_sgemm_summa")
+        return C
+
     def stats(self):
         st = MxStats()
         _ck(lib().mx_stats(self._ctx, ctypes.byref(st)))

@@ -349,3 +349,35 @@ def test_long_k_accumulation(eng):
     got = eng.dgemm(a, b)
     ref = a @ b
     assert rel_err(got, ref) < 1e-12
+
+
+def test_sgemm_summa_single_rank(eng):
+    # fp32 SUMMA path, 1x1 grid (comm_init done earlier in this module)
+    m, k, n = 400, 5000, 200
+    a = gen_matrix(m, k, seed=81, dtype=np.float32)
+    b = gen_matrix(k, n, seed=82, dtype=np.float32)
+    got = eng.sgemm_summa(m, k, n, a, b)
+    ref = a.astype(np.float64) @ b.astype(np.float64)
+    assert rel_err(got.astype(np.float64), ref) < 1e-4
+
+
+def test_error_paths(eng):
+    from marlin_amd.engine import EngineError
+    import ctypes
+    from marlin_amd import engine as E
+    lib = E.lib()
+    # mx_map binary op with null B -> MX_EINVAL
+    buf = np.zeros(4)
+    rc = lib.mx_map(eng._ctx, 0, 0, 4,
+                    buf.ctypes.data_as(ctypes.c_void_p), None, 0.0,
+                    buf.ctypes.data_as(ctypes.c_void_p))
+    assert rc == -4
+    # upload2d beyond capacity -> MX_EINVAL
+    d = eng.alloc(64)
+    rc = lib.mx_upload2d(eng._ctx, d, 8,
+                         buf.ctypes.data_as(ctypes.c_void_p), 8, 4, 8)
+    assert rc == -4
+    eng.free(d)
+    # negative dims -> MX_EDIM/EINVAL through the dgemm entry
+    rc = lib.mx_dgemm(eng._ctx, -1, 2, 2, None, None, None)
+    assert rc < 0

@@ -123,6 +123,17 @@ def lib():
     return _lib
 
 
+_LOG = os.environ.get("MARLIN_LOG")
+
+
+def _log(msg):
+    """stderr logger (SURVEY §5 observability plan): MARLIN_LOG=1 enables
+    per-call reports, =2 adds stage timings."""
+    if _LOG:
+        import sys
+        print(f"[marlin_amd] {msg}", file=sys.stderr, flush=True)
+
+
 def _ck(code, what=""):
     if code != 0:
         raise EngineError(code, what)
@@ -235,6 +246,11 @@ class Engine:
         _ck(lib().mx_dgemm(self._ctx, m, k, n, _fbuf(A, np.float64),
                            _fbuf(B, np.float64), _fbuf(C, np.float64)),
             "mx_dgemm")
+        if _LOG:
+            st = self.stats()
+            _log(f"dgemm {m}x{k}x{n}: gemm {st['gemm_ms']:.2f} ms "
+                 f"(h2d {st['h2d_ms']:.2f} ms, "
+                 f"{st['flops'] / max(st['gemm_ms'], 1e-9) / 1e9:.1f} TF/s)")
         return C
 
     def sgemm(self, A, B):

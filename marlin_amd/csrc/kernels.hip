@@ -58,14 +58,18 @@ DEVFN void glds16(const void* g, void* lds) {
 
 // ---------------------------------------------------------------------------
 // Main GEMM kernel. C[MxN] (+)= A[MxK] * B[KxN], col-major, padded pitches:
-// M,N multiples of 128, K a multiple of BK (host pads to 32 -> both BK
-// configs divide it). BETA == 1 accumulates into C (the SubMatrix.add
-// combiner, folded into the MFMA accumulator chain).
+// M a multiple of BM, N of 128, K of BK (the host pads to 128/16; the
+// launcher only selects BK=32 / BM=256 configs when the padded dims
+// divide). BETA == 1 accumulates into C (the SubMatrix.add combiner,
+// folded into the MFMA accumulator chain).
 //
-// Geometry template: BM=BN=128 fixed; BK in {16, 32}; NWAVES in {4, 8}
-// (wave grid 2 x NWAVES/2). BK=32/NWAVES=8 = one 512-thread block per CU
-// (128 KB LDS), half the barrier rate of BK=16/NWAVES=4 (two 256-thread
-// blocks per CU, 64 KB LDS each); both keep 2 waves/SIMD.
+// Geometry template: BN=128 fixed; wave grid WM x WN, wave tile
+// (BM/WM) x (128/WN). Production config (measured winner): BK=16,
+// BM=128, 2x2 waves = two 256-thread blocks per CU at 64 KB LDS each,
+// 2 waves/SIMD. Alternates kept for re-evaluation behind MARLIN_GEMM_CFG:
+// bk32 (512-thread, half barrier rate, 58.6 TF) and bm256 (256x128 tile,
+// half DMA-per-flop, 63.6 TF) — both measured slower than the default's
+// 67.9 TF at 20000^3.
 template <typename T, int BETA, int BK, int BM, int WM, int WN>
 __launch_bounds__(WM * WN * 64, 2)
 __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
@@ -128,22 +132,24 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
         : (BK / A_COLS_PER_GLDS / NWAVES);
     constexpr int B_COLS_PER_GLDS = (64 * E) / BK;
     constexpr int B_GLDS = BN / B_COLS_PER_GLDS / NWAVES;
-    // B swizzle parameters: chunk index within a column is XORed with
-    // (c >> BSH) & BMASK (fp64 only; see below)
+    // B swizzle parameters for fp64: chunk index within a column is
+    // XORed with (c >> BSH) & BMASK (fp32 uses its own masks; see below)
     constexpr int BMASK = BK / 2 - 1;
     constexpr int BSH = (BK == 16) ? 1 : 0;
 
-    // --- LDS bank-conflict swizzles (fp64 only; measured 8-way on the
-    // B-fragment reads without them — profiles/r01 PMC: conflict cycles
-    // were 88% of LDS cycles). glds forces a lane-linear LDS image, so the
+    // --- LDS bank-conflict swizzles (measured 8-way on the B-fragment
+    // reads without them — profiles/r01 PMC: conflict cycles were 88% of
+    // LDS cycles). glds forces a lane-linear LDS image, so the
     // swizzle is applied to the per-lane GLOBAL source address (guide
     // idiom); both XORs permute 16-byte chunks within one 128-byte global
     // run, so HBM coalescing is unchanged.
-    //   A image: column kk stores row-pair i at chunk i ^ (8*(kk&1))
+    //   fp64 A image: column kk stores row-pair i at chunk i ^ (8*(kk&1))
     //     -> a-read banks: lanes 0-15 distinct, lanes 16-31 shifted by 32.
-    //   B image: column c stores k-pair p at chunk p ^ ((c>>BSH)&BMASK)
+    //   fp64 B image: column c stores k-pair p at chunk p ^ ((c>>BSH)&BMASK)
     //     -> b-read banks: all 32 lanes of a ds_read_b64 group distinct
     //        (BK=16: column stride 128B; BK=32: column stride 256B).
+    //   fp32 (banks are mod 32 for b32 reads): A chunk ^ 4*(kk&1);
+    //     B chunk ^ (c>>1)&3 -> 2-way worst case (was 8-way).
     auto issue_tile = [&](int kt, int buf) {
         const int64_t kbase = (int64_t)kt * BK;
         if constexpr (BM * sizeof(T) >= 1024) {

@@ -167,6 +167,11 @@ def main():
         launches += st["gemm_launches"]
     # engine entries synchronize internally; no extra device sync needed
     dt = time.perf_counter() - t0
+    if os.environ.get("MARLIN_SUMMA_DEBUG") and world > 1:
+        st = eng.stats()
+        print(f"[rank {rank}] wall {dt*1e3:.1f} ms, last-step gemm "
+              f"{st['gemm_ms']:.1f} ms, comm {st['comm_ms']:.1f} ms, "
+              f"launches {st['gemm_launches']}", file=sys.stderr, flush=True)
     if dist:
         import torch
         tmax = torch.tensor([dt])

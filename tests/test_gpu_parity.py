@@ -381,3 +381,20 @@ def test_error_paths(eng):
     # negative dims -> MX_EDIM/EINVAL through the dgemm entry
     rc = lib.mx_dgemm(eng._ctx, -1, 2, 2, None, None, None)
     assert rc < 0
+
+
+def test_run_to_run_bitwise_deterministic(eng):
+    # fixed k-order accumulation + fixed reduction orders everywhere:
+    # identical inputs must give bitwise-identical results across runs
+    a = gen_matrix(700, 900, seed=91)
+    b = gen_matrix(900, 500, seed=92)
+    c1 = eng.dgemm(a, b)
+    c2 = eng.dgemm(a, b)
+    np.testing.assert_array_equal(c1, c2)
+    s1 = eng.sum(a)
+    s2 = eng.sum(a)
+    assert s1 == s2
+    x = gen_matrix(900, 1, seed=93)[:, 0]
+    y1 = eng.dgemv(a, x)
+    y2 = eng.dgemv(a, x)
+    np.testing.assert_array_equal(y1, y2)

@@ -89,14 +89,17 @@ int mx_sgemm_epilogue(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
 
 /* Distributed SUMMA (replaces the Spark shuffle route,
  * BlockMatrix.scala:161-186): every rank passes ITS OWN shards.
- * Layout (block distribution on the pr x pc grid):
- *   A_local: ceil-rows slab  [my_m x k]   (rows  owner: prow)
- *   B_local: ceil-cols slab  [k x my_n]   (cols  owner: pcol)
+ * Layout at rank (prow, pcol) of the pr x pc grid (see DESIGN.md §3;
+ * ceil splits, last slab ragged — DenseVecMatrix.scala:1262-1265
+ * semantics; my_x = mx_slab_len(x, parts, idx)):
+ *   A_local: [my_m x my_ka]  rows slab prow of m,  k-cols slab pcol of k
+ *   B_local: [my_kb x my_n]  k-rows slab prow of k, n-cols slab pcol of n
  *   C_local: [my_m x my_n]
- * my_m = slab_len(m, pr, prow), my_n = slab_len(n, pc, pcol) — ceil split,
- * last slab ragged (DenseVecMatrix.scala:1262-1265 semantics).
- * Panels of A (within grid rows) and B (within grid cols) are broadcast
- * per k-step over RCCL/xGMI, double-buffered against the MFMA stream. */
+ * Host-buffer entries take tight column-major shards; the engine pads
+ * internally. Per k-panel the owning column broadcasts its A panel in
+ * each grid row and the owning row its B panel in each grid column over
+ * RCCL/xGMI, double-buffered against the MFMA stream; each C shard has
+ * one owner, so no reduce exists. */
 int mx_dgemm_summa(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
                    const double* A_local, const double* B_local,
                    double* C_local);
@@ -152,7 +155,9 @@ int mx_gemm_device_ex(mx_ctx* ctx, int is_fp32, int beta_one, int64_t m,
                       int64_t ldc);
 
 /* SUMMA on device-resident local shards (bench hot loop: inputs already
- * in HBM when the timed region starts). */
+ * in HBM when the timed region starts). Shard pitches are the PADDED
+ * sizes: A_local pitch roundup(my_m,128) x my_ka cols; B_local pitch
+ * roundup(my_kb,16) x my_n cols; C_local pitch roundup(my_m,128). */
 int mx_dgemm_summa_device(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
                           const mx_dbuf* dA_local, const mx_dbuf* dB_local,
                           mx_dbuf* dC_local);

@@ -17,9 +17,6 @@ import torch.multiprocessing as mp
 from marlin_amd import engine as E
 from oracle import gen_matrix
 
-WORLD = 2
-
-
 def _summa_rank(rank, world, m, k, n, q, port):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
@@ -96,15 +93,17 @@ def _summa_rank(rank, world, m, k, n, q, port):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("mkn", [(96, 130, 64), (257, 99, 121)])
-def test_summa_dataflow_gloo_ws2(mkn):
+@pytest.mark.parametrize("world,mkn", [
+    (2, (96, 130, 64)), (2, (257, 99, 121)), (4, (150, 200, 120)),
+])
+def test_summa_dataflow_gloo(world, mkn):
     m, k, n = mkn
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29671 + (m % 100)
+    port = 29671 + (m % 100) + world
     procs = [ctx.Process(target=_summa_rank,
-                         args=(r, WORLD, m, k, n, q, port))
-             for r in range(WORLD)]
+                         args=(r, world, m, k, n, q, port))
+             for r in range(world)]
     for p in procs:
         p.start()
     for p in procs:

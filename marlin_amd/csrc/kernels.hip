@@ -280,6 +280,15 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
                 #pragma unroll
                 for (int nt = 0; nt < NT; nt++)
                     acc[mt][nt] = mfma_16x16x4(a[mt], b[nt], acc[mt][nt]);
+#ifdef MARLIN_SCHED_HINT
+            // scheduler hint: interleave next quad's DS reads between
+            // MFMAs, one read per two MFMAs (A/B experiment flag)
+            #pragma unroll
+            for (int g = 0; g < (MT + NT); g++) {
+                __builtin_amdgcn_sched_group_barrier(0x100, 1, 0);  // 1 DS read
+                __builtin_amdgcn_sched_group_barrier(0x008, (MT * NT) / (MT + NT) + 1, 0);  // MFMAs
+            }
+#endif
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         __builtin_amdgcn_s_barrier();        // readers done before overwrite

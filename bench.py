@@ -82,10 +82,18 @@ def main():
                    default=int(env.get("MARLIN_BENCH_NN", "0")))
     p.add_argument("--dtype", choices=["f64", "f32"],
                    default=env.get("MARLIN_BENCH_DTYPE", "f64"))
-    p.add_argument("--cpu-sample", type=int, default=4096)
+    # workload "epilogue": config 5's fused (A*B)^T + addC leg as its own
+    # timed line (fp32, device-resident), never the default
+    p.add_argument("--workload", choices=["gemm", "epilogue"],
+                   default=env.get("MARLIN_BENCH_WORKLOAD", "gemm"))
+    p.add_argument("--cpu-sample", type=int, default=12000)
     p.add_argument("--no-cpu-baseline", action="store_true")
     args = p.parse_args()
 
+    if args.workload == "epilogue":
+        args.dtype = "f32"
+        if args.n == 20000 and not (args.m or args.k or args.nn):
+            args.n = 40000          # config 5's quoted size
     m = args.m or args.n
     k = args.k or args.n
     n = args.nn or args.n
@@ -129,28 +137,61 @@ def main():
     mip, njp = roundup(mi, 128), roundup(nj, 128)
     kbi_p = roundup(kbi, 16)
 
+    kres = False
     if world > 1:
-        dA = eng.alloc(mip * kaj * elem)
-        dB = eng.alloc(kbi_p * nj * elem)
-        dC = eng.alloc(mip * njp * elem)
-        eng.fill_random(dA, mip * kaj, 0xA11CE + rank, fp32)
-        eng.fill_random(dB, kbi_p * nj, 0xB0B + rank, fp32)
+        # layout per CARMA splitMethod semantics (MTUtils.scala:150-175):
+        # kSplit==1 -> k-resident shards, zero steady-state xGMI traffic
+        # (config 4); else k-slabbed panel-broadcast SUMMA (config 3)
+        kres = eng_mod.summa_kresident(m, k, n, world)
+        if kres:
+            kp = roundup(k, 16)
+            dA = eng.alloc(mip * kp * elem)
+            dB = eng.alloc(kp * njp * elem)
+            dC = eng.alloc(mip * njp * elem)
+            eng.fill_random(dA, mip * kp, 0xA11CE + rank, fp32)
+            eng.fill_random(dB, kp * njp, 0xB0B + rank, fp32)
+            # zero-pad invariant: pad rows/cols must be zero
+            eng.zero_pad(dA, mip, kp, mip, mi, k, fp32)
+            eng.zero_pad(dB, kp, njp, kp, k, nj, fp32)
 
-        summa = eng.sgemm_summa_device if fp32 else eng.dgemm_summa_device
+            def step():
+                eng.gemm_summa_kres_device(m, k, n, dA, dB, dC, fp32)
+        else:
+            dA = eng.alloc(mip * kaj * elem)
+            dB = eng.alloc(kbi_p * nj * elem)
+            dC = eng.alloc(mip * njp * elem)
+            eng.fill_random(dA, mip * kaj, 0xA11CE + rank, fp32)
+            eng.fill_random(dB, kbi_p * nj, 0xB0B + rank, fp32)
+            eng.zero_pad(dA, mip, kaj, mip, mi, kaj, fp32)
+            eng.zero_pad(dB, kbi_p, nj, kbi_p, kbi, nj, fp32)
 
-        def step():
-            summa(m, k, n, dA, dB, dC)
+            summa = eng.sgemm_summa_device if fp32 else eng.dgemm_summa_device
+
+            def step():
+                summa(m, k, n, dA, dB, dC)
     else:
         mp, kp, np_ = roundup(m, 128), roundup(k, 16), roundup(n, 128)
         dA = eng.alloc(mp * kp * elem)
         dB = eng.alloc(kp * np_ * elem)
-        dC = eng.alloc(mp * np_ * elem)
         eng.fill_random(dA, mp * kp, 0xA11CE, fp32)
         eng.fill_random(dB, kp * np_, 0xB0B, fp32)
-        gem = eng.sgemm_device if fp32 else eng.dgemm_device
+        eng.zero_pad(dA, mp, kp, mp, m, k, fp32)
+        eng.zero_pad(dB, kp, np_, kp, k, n, fp32)
+        if args.workload == "epilogue":
+            # C is (A*B)^T + addC: n x m col-major, pitch np_
+            dC = eng.alloc(np_ * mp * elem)
+            dAdd = eng.alloc(np_ * mp * elem)
+            eng.fill_random(dAdd, np_ * mp, 0xADD, fp32)
 
-        def step():
-            gem(mp, kp, np_, dA, mp, dB, kp, dC, mp)
+            def step():
+                eng.sgemm_epilogue_device(mp, kp, np_, dA, mp, dB, kp,
+                                          dC, np_, dAdd)
+        else:
+            dC = eng.alloc(mp * np_ * elem)
+            gem = eng.sgemm_device if fp32 else eng.dgemm_device
+
+            def step():
+                gem(mp, kp, np_, dA, mp, dB, kp, dC, mp)
 
     # --- warmup / timed region ------------------------------------------
     for _ in range(args.warmup):
@@ -169,9 +210,10 @@ def main():
     dt = time.perf_counter() - t0
     if os.environ.get("MARLIN_SUMMA_DEBUG") and world > 1:
         st = eng.stats()
-        print(f"[rank {rank}] wall {dt*1e3:.1f} ms, last-step gemm "
-              f"{st['gemm_ms']:.1f} ms, comm {st['comm_ms']:.1f} ms, "
-              f"launches {st['gemm_launches']}", file=sys.stderr, flush=True)
+        print(f"[rank {rank}] layout {'kres' if kres else 'slab'}, wall "
+              f"{dt*1e3:.1f} ms, last-step gemm {st['gemm_ms']:.1f} ms, "
+              f"comm {st['comm_ms']:.1f} ms, launches "
+              f"{st['gemm_launches']}", file=sys.stderr, flush=True)
     if dist:
         import torch
         tmax = torch.tensor([dt])
@@ -200,6 +242,9 @@ def main():
             ach = (2.0 * m * k * n * args.steps) / (gemm_ms_acc / 1e3) / 1e12
         else:
             ach = None
+        wl = (f"sgemm_tn_epilogue_{m}x{k}x{n}_f32"
+              if args.workload == "epilogue" else
+              f"dgemm_{m}x{k}x{n}_{args.dtype}")
         # HBM traffic per launch: measured offline by rocprofv3 --pmc
         # FETCH_SIZE passes (tools_dev/rocpd_stats.py) and committed under
         # profiles/hbm_traffic.json keyed by workload; null when this
@@ -208,7 +253,7 @@ def main():
         try:
             tj = json.load(open(os.path.join(HERE, "profiles",
                                              "hbm_traffic.json")))
-            rec = tj.get(f"dgemm_{m}x{k}x{n}_{args.dtype}")
+            rec = tj.get(wl)
             if rec and world == 1:
                 traffic = rec["reads_bytes_per_launch"]
         except Exception:
@@ -221,9 +266,18 @@ def main():
             "frac": round(ach / (peak * world), 4) if ach else None,
             "traffic": traffic,
         }
+        if args.workload == "epilogue":
+            metric = "dense C=(AxB)^T+D TFLOP/s (fp32)"
+        else:
+            metric = ("dense C=AxB TFLOP/s (fp64)" if not fp32 else
+                      "dense C=AxB TFLOP/s (fp32)")
+        if world > 1:
+            par = (f"kres_grid_{pr}x{pc}" if kres else
+                   f"summa_grid_{pr}x{pc}")
+        else:
+            par = "single_gpu"
         out = {
-            "metric": "dense C=AxB TFLOP/s (fp64)" if not fp32 else
-                      "dense C=AxB TFLOP/s (fp32)",
+            "metric": metric,
             "value": round(tf, 3),
             "unit": "TFLOP/s",
             "n_gpus": world,
@@ -236,10 +290,9 @@ def main():
             "dtype": args.dtype,
             "data": "synthetic",
             "config": {
-                "workload": f"dgemm_{m}x{k}x{n}_{args.dtype}",
+                "workload": wl,
                 "m": m, "k": k, "n": n,
-                "parallelism": f"summa_grid_{pr}x{pc}" if world > 1 else
-                               "single_gpu",
+                "parallelism": par,
             },
             "roofline": roofline,
         }

@@ -107,6 +107,31 @@ int mx_sgemm_summa(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
                    const float* A_local, const float* B_local,
                    float* C_local);
 
+/* Distributed layout selector (CARMA semantics, MTUtils.scala:150-175):
+ * returns 1 when splitMethod(m,k,n,nranks) leaves k unsplit (kSplit==1),
+ * i.e. the k-RESIDENT layout applies — each rank keeps its A row-slab
+ * with ALL K columns and its B col-slab with ALL K rows in HBM
+ * (replicated across the other grid dimension) and the multiply is one
+ * local GEMM with ZERO steady-state xGMI traffic (BASELINE config 4:
+ * 50000x4096 · 4096x50000 on 8 GPUs). Returns 0 when the k-slabbed
+ * panel-broadcast SUMMA applies. */
+int mx_summa_kresident(int64_t m, int64_t k, int64_t n, int nranks);
+
+/* k-resident distributed multiply. Shard layout at rank (prow, pcol):
+ *   A_local: [my_m x K]   rows slab prow of m, ALL k columns
+ *   B_local: [K x my_n]   ALL k rows, n-cols slab pcol of n
+ *   C_local: [my_m x my_n]
+ * Host entries take tight col-major shards; the device entry (declared
+ * with the device-resident group below) takes padded pitches
+ * (A roundup(my_m,128) x K cols, B roundup(K,16) x my_n,
+ * C roundup(my_m,128)). No collective is issued. */
+int mx_dgemm_summa_kres(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
+                        const double* A_local, const double* B_local,
+                        double* C_local);
+int mx_sgemm_summa_kres(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
+                        const float* A_local, const float* B_local,
+                        float* C_local);
+
 /* Ceil slab split helper (exact reference blocking semantics). */
 int64_t mx_slab_len(int64_t total, int parts, int idx);
 int64_t mx_slab_off(int64_t total, int parts, int idx);
@@ -164,6 +189,40 @@ int mx_dgemm_summa_device(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
 int mx_sgemm_summa_device(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
                           const mx_dbuf* dA_local, const mx_dbuf* dB_local,
                           mx_dbuf* dC_local);
+/* k-resident layout on device-resident shards (see mx_summa_kresident). */
+int mx_gemm_summa_kres_device(mx_ctx* ctx, int is_fp32, int64_t m, int64_t k,
+                              int64_t n, const mx_dbuf* dA_local,
+                              const mx_dbuf* dB_local, mx_dbuf* dC_local);
+
+/* Restore the zero-pad invariant of a rows_total x cols_total padded
+ * image (pitch ld) whose logical content is m x n: pads outside m x n
+ * are zeroed (used after mx_fill_random of a whole padded buffer). */
+int mx_zero_pad(mx_ctx* ctx, mx_dbuf* buf, int64_t rows_total,
+                int64_t cols_total, int64_t ld, int64_t m, int64_t n,
+                int is_fp32);
+
+/* Download starting at a byte offset (one COLUMN of a padded col-major
+ * image: off = j * pitch * elem). */
+int mx_download_off(mx_ctx* ctx, void* dst, const mx_dbuf* src,
+                    int64_t off_bytes, int64_t bytes);
+/* Pitched download starting at a byte offset (one ROW of a col-major
+ * image: off = i * elem, m = 1, n = cols, pitch = ld). */
+int mx_download2d_off(mx_ctx* ctx, void* dst, const mx_dbuf* src,
+                      int64_t off_bytes, int64_t pitch_elems, int64_t m,
+                      int64_t n, int elem);
+
+/* Device-resident fused-epilogue GEMM (config 5 timed leg): C[n x m] =
+ * (A*B)^T (+ addC), padded dims, ldc = padded-n pitch of C/addC;
+ * dAdd may be NULL. */
+int mx_sgemm_epilogue_device(mx_ctx* ctx, int64_t m, int64_t k, int64_t n,
+                             const mx_dbuf* dA, int64_t lda,
+                             const mx_dbuf* dB, int64_t ldb, mx_dbuf* dC,
+                             int64_t ldc, const mx_dbuf* dAdd);
+
+/* Failure-injection probe: issues an invalid RCCL collective
+ * on the world communicator; a healthy engine surfaces MX_ERCCL (no
+ * abort) and keeps serving calls afterwards. */
+int mx_test_rccl_error(mx_ctx* ctx);
 
 /* ---- elementwise / reduction / transpose (BlockMatrix epilogue ops,
  * BlockMatrix.scala:344-523; DenseVecMatrix.scala scalar ops) ---------- */

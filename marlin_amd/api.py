@@ -136,6 +136,12 @@ class DenseVecMatrix:
         m, k, n = self.numRows(), self.numCols(), other.numCols()
         bsize = broadcast_threshold * 1024 * 1024 // 8
         if self._dev is not None and other._dev is not None:
+            # Deliberate divergence from the reference dispatch: cached
+            # (device-resident) operands return a device-resident
+            # DenseVecMatrix instead of a BlockMatrix — re-blocking the
+            # result would force a D2H round trip, defeating cache().
+            # Call .toBlockMatrix(r, c) on the result to get the
+            # reference's block layout when needed.
             cdev = self._engine().gemm_dd(self._dev, other._dev)
             return DenseVecMatrix(None, self._eng, _dev=cdev)
         c = self._engine().dgemm(self._host(), other._host())

@@ -603,6 +603,17 @@ int mxk_gemm(int is_fp32, int beta_one,
              int64_t M, int64_t N, int64_t K,
              const void* A, int64_t lda, const void* B, int64_t ldb,
              void* C, int64_t ldc, hipStream_t stream) {
+    if (M < 0 || N < 0 || K < 0) return -4;
+    if (M == 0 || N == 0) return 0;            // empty tile: nothing to do
+    if (K == 0) {                              // C = 0 (or unchanged if +=)
+        if (!beta_one) {
+            size_t es = is_fp32 ? 4 : 8;
+            hipError_t e = hipMemset2DAsync(C, (size_t)ldc * es, 0,
+                                            (size_t)M * es, (size_t)N, stream);
+            if (e != hipSuccess) return -2;
+        }
+        return 0;
+    }
     if (M % 128 || N % 128 || K % 16) return -4;
     int nbn = (int)(N / 128);
     static const char* bandenv = getenv("MARLIN_GEMM_BAND");

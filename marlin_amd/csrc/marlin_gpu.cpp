@@ -131,6 +131,16 @@ const char* mx_strerror(int code) {
   }
 }
 
+// destroys whatever streams/events a partially-initialised ctx created
+static void ctx_teardown(mx_ctx* c) {
+  for (int i = 0; i < 16; i++)
+    if (c->ev[i]) (void)hipEventDestroy(c->ev[i]);
+  if (c->s_gemm) (void)hipStreamDestroy(c->s_gemm);
+  if (c->s_copy) (void)hipStreamDestroy(c->s_copy);
+  if (c->s_comm) (void)hipStreamDestroy(c->s_comm);
+  delete c;
+}
+
 int mx_init(mx_ctx** out, int device) {
   if (!out) return MX_EINVAL;
   int ndev = 0;
@@ -141,11 +151,11 @@ int mx_init(mx_ctx** out, int device) {
   if (hipStreamCreate(&c->s_gemm) != hipSuccess ||
       hipStreamCreate(&c->s_copy) != hipSuccess ||
       hipStreamCreate(&c->s_comm) != hipSuccess) {
-    delete c;
+    ctx_teardown(c);
     return MX_EHIP;
   }
   for (int i = 0; i < 16; i++)
-    if (hipEventCreate(&c->ev[i]) != hipSuccess) { delete c; return MX_EHIP; }
+    if (hipEventCreate(&c->ev[i]) != hipSuccess) { ctx_teardown(c); return MX_EHIP; }
   *out = c;
   return MX_OK;
 }
@@ -278,6 +288,49 @@ int mx_fill_random(mx_ctx* c, mx_dbuf* buf, int64_t n_elems, uint64_t seed,
                            c->s_gemm);
   if (rc) return rc;
   HIP_OK(hipStreamSynchronize(c->s_gemm));
+  return MX_OK;
+}
+
+// Zero the pad region of an m x n logical image inside a rows_total x
+// cols_total padded buffer (pitch ld) — restores the zero-pad invariant
+// after a whole-buffer fill.
+int mx_zero_pad(mx_ctx* c, mx_dbuf* buf, int64_t rows_total,
+                int64_t cols_total, int64_t ld, int64_t m, int64_t n,
+                int is_fp32) {
+  if (!c || !buf) return MX_EINVAL;
+  HIP_OK(hipSetDevice(c->device));
+  if (mxk_zero_pad(is_fp32, buf->ptr, rows_total, cols_total, ld, m, n,
+                   c->s_gemm))
+    return MX_EHIP;
+  HIP_OK(hipStreamSynchronize(c->s_gemm));
+  return MX_OK;
+}
+
+// Download starting at a byte offset into the device buffer (e.g. one
+// column of a padded col-major image: offset = j * pitch * elem).
+int mx_download_off(mx_ctx* c, void* dst, const mx_dbuf* src,
+                    int64_t off_bytes, int64_t bytes) {
+  if (!c || !dst || !src || off_bytes < 0 || bytes <= 0 ||
+      off_bytes + bytes > src->bytes)
+    return MX_EINVAL;
+  HIP_OK(hipSetDevice(c->device));
+  HIP_OK(hipMemcpy(dst, (const char*)src->ptr + off_bytes, (size_t)bytes,
+                   hipMemcpyDeviceToHost));
+  return MX_OK;
+}
+
+// 2D pitched download starting at a byte offset (e.g. one ROW of a
+// col-major image: offset = i * elem, m = 1, n = cols, pitch = ld).
+int mx_download2d_off(mx_ctx* c, void* dst, const mx_dbuf* src,
+                      int64_t off_bytes, int64_t pitch_elems, int64_t m,
+                      int64_t n, int elem) {
+  if (!c || !dst || !src || m <= 0 || n <= 0 || off_bytes < 0)
+    return MX_EINVAL;
+  HIP_OK(hipSetDevice(c->device));
+  HIP_OK(hipMemcpy2D(dst, (size_t)(m * elem),
+                     (const char*)src->ptr + off_bytes,
+                     (size_t)(pitch_elems * elem), (size_t)(m * elem),
+                     (size_t)n, hipMemcpyDeviceToHost));
   return MX_OK;
 }
 
@@ -538,6 +591,22 @@ extern "C" int mx_plan_panels(int64_t K, int pr, int pc, int64_t kb_max,
   return (int)v.size();
 }
 
+// RAII pool for the per-panel timing events: destroyed on EVERY exit path
+// (early error returns included), so failed SUMMA calls leak nothing.
+struct ev_pool {
+  std::vector<hipEvent_t> v;
+  hipEvent_t mk() {
+    hipEvent_t e = nullptr;
+    (void)hipEventCreate(&e);
+    v.push_back(e);
+    return e;
+  }
+  ~ev_pool() {
+    for (hipEvent_t e : v)
+      if (e) (void)hipEventDestroy(e);
+  }
+};
+
 static int summa_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
                         int64_t n, const void* dA, const void* dB, void* dC) {
   if (!c->have_comm) return MX_ENOCOMM;
@@ -545,11 +614,15 @@ static int summa_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
   const ncclDataType_t nty = is_fp32 ? ncclFloat32 : ncclFloat64;
   const int64_t mi = mx_slab_len(m, c->pr, c->prow);
   const int64_t nj = mx_slab_len(n, c->pc, c->pcol);
-  const int64_t kaj = mx_slab_len(k, c->pc, c->pcol);   // A k-cols here
   const int64_t kbi = mx_slab_len(k, c->pr, c->prow);   // B k-rows here
   const int64_t ka_off = mx_slab_off(k, c->pc, c->pcol);
   const int64_t kb_off = mx_slab_off(k, c->pr, c->prow);
   const int64_t mip = round_up(mi, 128), njp = round_up(nj, 128);
+  // Empty local shard (e.g. m < pr on tiny inputs): the reference path
+  // handles any size, so this rank must still take part in every panel
+  // broadcast (counts are uniform per row/col comm: mip is shared by the
+  // whole grid row, njp by the whole grid column) but launches no GEMM.
+  const bool has_tile = mip > 0 && njp > 0;
   // local shard pitches ARE the padded sizes (bench fills them that way;
   // host entry packs them that way)
   // panel width: overlap granularity of the comm/MFMA pipeline
@@ -566,8 +639,10 @@ static int summa_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
   int rc;
   const int64_t kbp_max = round_up(kb_max, 16);
   for (int b = 0; b < 2; b++) {
-    if ((rc = ensure(c, &c->wsPA[b], mip * kbp_max * elem))) return rc;
-    if ((rc = ensure(c, &c->wsPB[b], kbp_max * njp * elem))) return rc;
+    if (mip > 0 && (rc = ensure(c, &c->wsPA[b], mip * kbp_max * elem)))
+      return rc;
+    if (njp > 0 && (rc = ensure(c, &c->wsPB[b], kbp_max * njp * elem)))
+      return rc;
   }
 
   // ev[8+b]: gemm done reading panel buffer b; ev[12+b]: panel b ready
@@ -577,13 +652,7 @@ static int summa_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
   // deferred per-panel timing: events recorded in-loop, synchronised
   // only AFTER both streams drain (an in-loop sync would serialise the
   // comm/GEMM double-buffer pipeline)
-  std::vector<hipEvent_t> tev;
-  auto mkev = [&]() {
-    hipEvent_t e;
-    (void)hipEventCreate(&e);
-    tev.push_back(e);
-    return e;
-  };
+  ev_pool tev;
   std::vector<std::pair<hipEvent_t, hipEvent_t>> gemm_tv, comm_tv;
 
   for (size_t p = 0; p < panels.size(); p++) {
@@ -598,17 +667,19 @@ static int summa_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
     // is done, then pack (root) and broadcast.
     HIP_OK(hipStreamWaitEvent(c->s_comm, c->ev[8 + buf], 0));
     if (kbp != kb) {
-      HIP_OK(hipMemsetAsync(pa, 0, (size_t)(mip * kbp * elem), c->s_comm));
-      HIP_OK(hipMemsetAsync(pb, 0, (size_t)(kbp * njp * elem), c->s_comm));
+      if (mip > 0)
+        HIP_OK(hipMemsetAsync(pa, 0, (size_t)(mip * kbp * elem), c->s_comm));
+      if (njp > 0)
+        HIP_OK(hipMemsetAsync(pb, 0, (size_t)(kbp * njp * elem), c->s_comm));
     }
-    if (c->pcol == pan.rootA) {
+    if (c->pcol == pan.rootA && mip > 0) {
       // A panel: k-cols [k0-ka_off, k1-ka_off) of A_local (pitch mip) are
       // contiguous -> strided copy into packed panel (pitch mip, kb cols)
       const char* src = (const char*)dA + (pan.k0 - ka_off) * mip * elem;
       HIP_OK(hipMemcpyAsync(pa, src, (size_t)(mip * kb * elem),
                             hipMemcpyDeviceToDevice, c->s_comm));
     }
-    if (c->prow == pan.rootB) {
+    if (c->prow == pan.rootB && njp > 0 && kbi > 0) {
       // B panel: k-rows [k0-kb_off, k1-kb_off) of B_local (pitch kbi_p):
       // strided 2D copy into packed pitch kbp
       const int64_t kbi_p = round_up(kbi, 16);
@@ -617,7 +688,7 @@ static int summa_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
                               nj, hipMemcpyDeviceToDevice, c->s_comm));
     }
     if (c->nranks > 1) {
-      hipEvent_t c0 = mkev(), c1 = mkev();
+      hipEvent_t c0 = tev.mk(), c1 = tev.mk();
       HIP_OK(hipEventRecord(c0, c->s_comm));
       RCCL_OK(ncclGroupStart());
       RCCL_OK(ncclBroadcast(pa, pa, (size_t)(mip * kbp), nty, pan.rootA,
@@ -632,15 +703,17 @@ static int summa_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
 
     // gemm stream: wait for the panel, accumulate
     HIP_OK(hipStreamWaitEvent(c->s_gemm, c->ev[12 + buf], 0));
-    int beta = p == 0 ? 0 : 1;
-    hipEvent_t g0 = mkev(), g1 = mkev();
-    HIP_OK(hipEventRecord(g0, c->s_gemm));
-    rc = mxk_gemm(is_fp32, beta, mip, njp, kbp, pa, mip, pb, kbp, dC, mip,
-                  c->s_gemm);
-    if (rc) return rc == -4 ? MX_EINVAL : MX_EHIP;
-    c->st.gemm_launches += 1;
-    HIP_OK(hipEventRecord(g1, c->s_gemm));
-    gemm_tv.push_back({g0, g1});
+    if (has_tile) {
+      int beta = p == 0 ? 0 : 1;
+      hipEvent_t g0 = tev.mk(), g1 = tev.mk();
+      HIP_OK(hipEventRecord(g0, c->s_gemm));
+      rc = mxk_gemm(is_fp32, beta, mip, njp, kbp, pa, mip, pb, kbp, dC, mip,
+                    c->s_gemm);
+      if (rc) return rc == -4 ? MX_EINVAL : MX_EHIP;
+      c->st.gemm_launches += 1;
+      HIP_OK(hipEventRecord(g1, c->s_gemm));
+      gemm_tv.push_back({g0, g1});
+    }
     HIP_OK(hipEventRecord(c->ev[8 + buf], c->s_gemm));
   }
   HIP_OK(hipStreamSynchronize(c->s_gemm));
@@ -655,8 +728,114 @@ static int summa_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
     (void)hipEventElapsedTime(&ms, pr2.first, pr2.second);
     c->st.comm_ms += ms;
   }
-  for (hipEvent_t e : tev) (void)hipEventDestroy(e);
   return MX_OK;
+}
+
+// ---------------------------------------------------------------------------
+// k-resident distributed layout (BASELINE config 4).
+//
+// CARMA splitMethod (MTUtils.scala:150-175) halves the LARGEST of m,k,n
+// per step; for tall-skinny x short-fat shapes (50000x4096 · 4096x50000 on
+// 8 GPUs) it never splits k -> kSplit == 1. The MI355X layout mirrors that
+// choice: each rank keeps its A row-slab with ALL K columns and its B
+// col-slab with ALL K rows resident in HBM (replicated across the other
+// grid dimension), so the multiply is ONE local MFMA GEMM with ZERO
+// steady-state xGMI traffic — no panel broadcasts at all.
+static void split_method_c(int64_t m, int64_t k, int64_t n, int cores,
+                           int* ms, int* ks, int* ns) {
+  // exact MTUtils.scala:150-175 semantics (n tested first, then m, else k)
+  *ms = *ks = *ns = 1;
+  int64_t _m = m, _k = k, _n = n;
+  int c = cores;
+  while (c > 1 && _m > 1 && _k > 1 && _n > 1) {
+    if (_n >= _k && _n >= _m) { *ns *= 2; _n /= 2; }
+    else if (_m >= _k && _m >= _n) { *ms *= 2; _m /= 2; }
+    else { *ks *= 2; _k /= 2; }
+    c /= 2;
+  }
+}
+
+int mx_summa_kresident(int64_t m, int64_t k, int64_t n, int nranks) {
+  if (nranks <= 1) return 0;
+  int ms, ks, ns;
+  split_method_c(m, k, n, nranks, &ms, &ks, &ns);
+  return ks == 1 ? 1 : 0;
+}
+
+static int summa_kres_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
+                             int64_t n, const void* dA, const void* dB,
+                             void* dC) {
+  if (!c->have_comm) return MX_ENOCOMM;
+  const int64_t elem = is_fp32 ? 4 : 8;
+  const int64_t mi = mx_slab_len(m, c->pr, c->prow);
+  const int64_t nj = mx_slab_len(n, c->pc, c->pcol);
+  const int64_t mip = round_up(mi, 128), njp = round_up(nj, 128);
+  const int64_t kp = round_up(k, 16);
+  c->st = {};
+  c->st.flops = 2.0 * m * k * n;
+  c->st.bytes_moved = (double)elem * (m * k + k * n + m * n);
+  if (mip == 0 || njp == 0) return MX_OK;  // empty shard: nothing to do
+  HIP_OK(hipSetDevice(c->device));
+  HIP_OK(hipEventRecord(c->ev[0], c->s_gemm));
+  int rc = mxk_gemm(is_fp32, 0, mip, njp, kp, dA, mip, dB, kp, dC, mip,
+                    c->s_gemm);
+  if (rc) return rc == -4 ? MX_EINVAL : MX_EHIP;
+  HIP_OK(hipEventRecord(c->ev[1], c->s_gemm));
+  HIP_OK(hipEventSynchronize(c->ev[1]));
+  float ms = 0;
+  HIP_OK(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
+  c->st.gemm_ms += ms;
+  c->st.gemm_launches += 1;
+  return MX_OK;
+}
+
+int mx_gemm_summa_kres_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
+                              int64_t n, const mx_dbuf* dA, const mx_dbuf* dB,
+                              mx_dbuf* dC) {
+  if (!c || !dA || !dB || !dC) return MX_EINVAL;
+  return summa_kres_device(c, is_fp32, m, k, n, dA->ptr, dB->ptr, dC->ptr);
+}
+
+// host-buffer k-resident entry: shards are A_local mi x K, B_local K x nj
+// (tight col-major); engine pads and runs the single local GEMM.
+static int summa_kres_host(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
+                           int64_t n, const void* A, const void* B, void* C) {
+  if (!c || !A || !B || !C) return MX_EINVAL;
+  if (!c->have_comm) return MX_ENOCOMM;
+  const int64_t elem = is_fp32 ? 4 : 8;
+  const int64_t mi = mx_slab_len(m, c->pr, c->prow);
+  const int64_t nj = mx_slab_len(n, c->pc, c->pcol);
+  const int64_t mip = round_up(mi, 128), njp = round_up(nj, 128);
+  const int64_t kp = round_up(k, 16);
+  if (mi == 0 || nj == 0) return MX_OK;
+  int rc;
+  HIP_OK(hipSetDevice(c->device));
+  if ((rc = ensure(c, &c->wsA, mip * kp * elem))) return rc;
+  if ((rc = ensure(c, &c->wsB, kp * njp * elem))) return rc;
+  if ((rc = ensure(c, &c->wsC, mip * njp * elem))) return rc;
+  if (mi != mip || k != kp)
+    HIP_OK(hipMemset(c->wsA.ptr, 0, (size_t)(mip * kp * elem)));
+  if (k != kp || nj != njp)
+    HIP_OK(hipMemset(c->wsB.ptr, 0, (size_t)(kp * njp * elem)));
+  HIP_OK(hipMemcpy2D(c->wsA.ptr, mip * elem, A, mi * elem, mi * elem, k,
+                     hipMemcpyHostToDevice));
+  HIP_OK(hipMemcpy2D(c->wsB.ptr, kp * elem, B, k * elem, k * elem, nj,
+                     hipMemcpyHostToDevice));
+  if ((rc = summa_kres_device(c, is_fp32, m, k, n, c->wsA.ptr, c->wsB.ptr,
+                              c->wsC.ptr)))
+    return rc;
+  HIP_OK(hipMemcpy2D(C, mi * elem, c->wsC.ptr, mip * elem, mi * elem, nj,
+                     hipMemcpyDeviceToHost));
+  return MX_OK;
+}
+
+int mx_dgemm_summa_kres(mx_ctx* c, int64_t m, int64_t k, int64_t n,
+                        const double* A, const double* B, double* C) {
+  return summa_kres_host(c, 0, m, k, n, A, B, C);
+}
+int mx_sgemm_summa_kres(mx_ctx* c, int64_t m, int64_t k, int64_t n,
+                        const float* A, const float* B, float* C) {
+  return summa_kres_host(c, 1, m, k, n, A, B, C);
 }
 
 int mx_dgemm_summa_device(mx_ctx* c, int64_t m, int64_t k, int64_t n,
@@ -687,19 +866,22 @@ static int summa_host(mx_ctx* c, int is_fp32, int64_t m, int64_t k, int64_t n,
   if ((rc = ensure(c, &c->wsA, mip * (kaj > 0 ? kaj : 1) * elem))) return rc;
   if ((rc = ensure(c, &c->wsB, kbi_p * (nj > 0 ? nj : 1) * elem))) return rc;
   if ((rc = ensure(c, &c->wsC, mip * njp * elem))) return rc;
-  if (mi != mip)
+  if (mi != mip && kaj > 0)
     HIP_OK(hipMemset(c->wsA.ptr, 0, (size_t)(mip * kaj * elem)));
-  if (kbi != kbi_p)
+  if (kbi != kbi_p && nj > 0)
     HIP_OK(hipMemset(c->wsB.ptr, 0, (size_t)(kbi_p * nj * elem)));
-  HIP_OK(hipMemcpy2D(c->wsA.ptr, mip * elem, A, mi * elem, mi * elem, kaj,
-                     hipMemcpyHostToDevice));
-  HIP_OK(hipMemcpy2D(c->wsB.ptr, kbi_p * elem, B, kbi * elem, kbi * elem, nj,
-                     hipMemcpyHostToDevice));
+  if (mi > 0 && kaj > 0)
+    HIP_OK(hipMemcpy2D(c->wsA.ptr, mip * elem, A, mi * elem, mi * elem, kaj,
+                       hipMemcpyHostToDevice));
+  if (kbi > 0 && nj > 0)
+    HIP_OK(hipMemcpy2D(c->wsB.ptr, kbi_p * elem, B, kbi * elem, kbi * elem,
+                       nj, hipMemcpyHostToDevice));
   if ((rc = summa_device(c, is_fp32, m, k, n, c->wsA.ptr, c->wsB.ptr,
                          c->wsC.ptr)))
     return rc;
-  HIP_OK(hipMemcpy2D(C, mi * elem, c->wsC.ptr, mip * elem, mi * elem, nj,
-                     hipMemcpyDeviceToHost));
+  if (mi > 0 && nj > 0)
+    HIP_OK(hipMemcpy2D(C, mi * elem, c->wsC.ptr, mip * elem, mi * elem, nj,
+                       hipMemcpyDeviceToHost));
   return MX_OK;
 }
 
@@ -820,5 +1002,48 @@ int mx_dgemv_device(mx_ctx* c, int64_t m, int64_t n, const mx_dbuf* dA,
 int mx_stats(mx_ctx* c, mx_stats_t* out) {
   if (!c || !out) return MX_EINVAL;
   *out = c->st;
+  return MX_OK;
+}
+
+// Device-resident fused-epilogue GEMM (BASELINE config 5 timed leg):
+// C[n x m] = (A*B)^T (+ addC), all buffers already in HBM, padded dims
+// (m,n mult of 128, k of 16). ldc is the padded n pitch of C/addC.
+int mx_sgemm_epilogue_device(mx_ctx* c, int64_t m, int64_t k, int64_t n,
+                             const mx_dbuf* dA, int64_t lda,
+                             const mx_dbuf* dB, int64_t ldb, mx_dbuf* dC,
+                             int64_t ldc, const mx_dbuf* dAdd) {
+  if (!c || !dA || !dB || !dC) return MX_EINVAL;
+  if (m % 128 || n % 128 || k % 16) return MX_EINVAL;
+  c->st = {};
+  c->st.flops = 2.0 * m * k * n;
+  c->st.bytes_moved = 4.0 * (m * k + k * n + m * n + (dAdd ? m * n : 0));
+  HIP_OK(hipSetDevice(c->device));
+  HIP_OK(hipEventRecord(c->ev[0], c->s_gemm));
+  int rc = mxk_sgemm_tn_epilogue(m, n, k, (const float*)dA->ptr, lda,
+                                 (const float*)dB->ptr, ldb, (float*)dC->ptr,
+                                 ldc, dAdd ? (const float*)dAdd->ptr : nullptr,
+                                 c->s_gemm);
+  if (rc) return rc == -4 ? MX_EINVAL : MX_EHIP;
+  HIP_OK(hipEventRecord(c->ev[1], c->s_gemm));
+  HIP_OK(hipEventSynchronize(c->ev[1]));
+  float ms = 0;
+  HIP_OK(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
+  c->st.gemm_ms += ms;
+  c->st.gemm_launches += 1;
+  return MX_OK;
+}
+
+// Failure-injection probe (RCCL error codes surfaced through
+// the ABI, never aborts): issues an INVALID collective (null buffer) on
+// the world communicator and returns the surfaced error. A healthy
+// engine returns MX_ERCCL here and keeps working afterwards.
+int mx_test_rccl_error(mx_ctx* c) {
+  if (!c) return MX_EINVAL;
+  if (!c->have_comm) return MX_ENOCOMM;
+  HIP_OK(hipSetDevice(c->device));
+  ncclResult_t e = ncclBroadcast(nullptr, nullptr, 16, ncclFloat64, 0,
+                                 c->world, c->s_comm);
+  if (e != ncclSuccess) return MX_ERCCL;  // expected: invalid argument
+  HIP_OK(hipStreamSynchronize(c->s_comm));
   return MX_OK;
 }

@@ -87,6 +87,22 @@ def _load():
                                            vp, i64]),
         "mx_dgemm_summa_device": (ctypes.c_int, [vp, i64, i64, i64, vp, vp, vp]),
         "mx_sgemm_summa_device": (ctypes.c_int, [vp, i64, i64, i64, vp, vp, vp]),
+        "mx_summa_kresident": (ctypes.c_int, [i64, i64, i64, ctypes.c_int]),
+        "mx_dgemm_summa_kres": (ctypes.c_int, [vp, i64, i64, i64, P(dbl),
+                                               P(dbl), P(dbl)]),
+        "mx_sgemm_summa_kres": (ctypes.c_int, [vp, i64, i64, i64, P(flt),
+                                               P(flt), P(flt)]),
+        "mx_gemm_summa_kres_device": (ctypes.c_int, [vp, ctypes.c_int, i64,
+                                                     i64, i64, vp, vp, vp]),
+        "mx_zero_pad": (ctypes.c_int, [vp, vp, i64, i64, i64, i64, i64,
+                                       ctypes.c_int]),
+        "mx_download_off": (ctypes.c_int, [vp, vp, vp, i64, i64]),
+        "mx_download2d_off": (ctypes.c_int, [vp, vp, vp, i64, i64, i64, i64,
+                                             ctypes.c_int]),
+        "mx_sgemm_epilogue_device": (ctypes.c_int, [vp, i64, i64, i64, vp,
+                                                    i64, vp, i64, vp, i64,
+                                                    vp]),
+        "mx_test_rccl_error": (ctypes.c_int, [vp]),
         "mx_map": (ctypes.c_int, [vp, ctypes.c_int, ctypes.c_int, i64, vp,
                                   vp, dbl, vp]),
         "mx_sum": (ctypes.c_int, [vp, ctypes.c_int, i64, vp, P(dbl)]),
@@ -163,6 +179,13 @@ def plan_panels(K, pr, pc, kb_max=4096):
 def grid_shape(nranks):
     return {8: (4, 2), 4: (2, 2), 2: (2, 1), 1: (1, 1)}.get(nranks,
                                                             (nranks, 1))
+
+
+def summa_kresident(m, k, n, nranks):
+    """True when CARMA splitMethod (MTUtils.scala:150-175) leaves k
+    unsplit for this job size -> the k-resident distributed layout
+    applies (zero steady-state xGMI traffic; BASELINE config 4)."""
+    return bool(lib().mx_summa_kresident(m, k, n, nranks))
 
 
 def _fbuf(a, dtype):
@@ -380,6 +403,59 @@ class Engine:
                                  _fbuf(B_local, np.float32),
                                  _fbuf(C, np.float32)), "mx_sgemm_summa")
         return C
+
+    def dgemm_summa_kres(self, m, k, n, A_local, B_local):
+        """k-resident distributed multiply (config-4 layout): A_local is
+        this rank's row slab with ALL k columns, B_local its col slab
+        with ALL k rows; one local GEMM, zero collectives."""
+        A_local = np.asfortranarray(A_local, dtype=np.float64)
+        B_local = np.asfortranarray(B_local, dtype=np.float64)
+        pr, pc, prow, pcol = self.grid()
+        mi = slab_len(m, pr, prow)
+        nj = slab_len(n, pc, pcol)
+        C = np.empty((mi, nj), dtype=np.float64, order="F")
+        _ck(lib().mx_dgemm_summa_kres(self._ctx, m, k, n,
+                                      _fbuf(A_local, np.float64),
+                                      _fbuf(B_local, np.float64),
+                                      _fbuf(C, np.float64)),
+            "mx_dgemm_summa_kres")
+        return C
+
+    def gemm_summa_kres_device(self, m, k, n, dA, dB, dC, fp32=False):
+        _ck(lib().mx_gemm_summa_kres_device(self._ctx, 1 if fp32 else 0,
+                                            m, k, n, dA, dB, dC),
+            "mx_gemm_summa_kres_device")
+
+    def zero_pad(self, dbuf, rows_total, cols_total, ld, m, n, fp32=False):
+        _ck(lib().mx_zero_pad(self._ctx, dbuf, rows_total, cols_total, ld,
+                              m, n, 1 if fp32 else 0), "mx_zero_pad")
+
+    def download_off(self, host_arr, dbuf, off_bytes, nbytes):
+        _ck(lib().mx_download_off(self._ctx,
+                                  host_arr.ctypes.data_as(ctypes.c_void_p),
+                                  dbuf, off_bytes, nbytes), "mx_download_off")
+
+    def download_row(self, dbuf, row, pitch_elems, ncols, fp32=False):
+        """One row of a padded col-major device image."""
+        dt = np.float32 if fp32 else np.float64
+        elem = 4 if fp32 else 8
+        out = np.empty(ncols, dtype=dt)
+        _ck(lib().mx_download2d_off(self._ctx,
+                                    out.ctypes.data_as(ctypes.c_void_p),
+                                    dbuf, row * elem, pitch_elems, 1, ncols,
+                                    elem), "mx_download2d_off")
+        return out
+
+    def sgemm_epilogue_device(self, m, k, n, dA, lda, dB, ldb, dC, ldc,
+                              dAdd=None):
+        _ck(lib().mx_sgemm_epilogue_device(self._ctx, m, k, n, dA, lda, dB,
+                                           ldb, dC, ldc, dAdd),
+            "mx_sgemm_epilogue_device")
+
+    def test_rccl_error(self):
+        """Failure-injection probe: returns the raw code (expected
+        MX_ERCCL = -6 for the injected invalid collective)."""
+        return lib().mx_test_rccl_error(self._ctx)
 
     def stats(self):
         st = MxStats()

@@ -37,7 +37,9 @@ def repeat_by_row(matrix, times):
         blocks = {}
         for (i, j), b in matrix._blocks.items():
             for t in range(times):
-                blocks[(i, j + t * nbc)] = b
+                # independent copies: repeated blocks must not alias one
+                # ndarray (in-place ops on one copy would corrupt all)
+                blocks[(i, j + t * nbc)] = b if t == 0 else b.copy()
         return BlockMatrix(blocks, matrix.numRows(),
                            matrix.numCols() * times, engine=matrix._eng)
     a = matrix.toBreeze()
@@ -55,7 +57,7 @@ def repeat_by_column(matrix, times):
         blocks = {}
         for (i, j), b in matrix._blocks.items():
             for t in range(times):
-                blocks[(i + t * nbr, j)] = b
+                blocks[(i + t * nbr, j)] = b if t == 0 else b.copy()
         return BlockMatrix(blocks, matrix.numRows() * times,
                            matrix.numCols(), engine=matrix._eng)
     a = matrix.toBreeze()

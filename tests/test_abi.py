@@ -68,11 +68,91 @@ def test_panel_plan(K, pr, pc):
         assert 0 <= ra < pc and 0 <= rb < pr
 
 
+def test_summa_kresident_layout_select():
+    # CARMA splitMethod semantics (MTUtils.scala:150-175): config 4
+    # (50000x4096 . 4096x50000) never splits k on <=8 ranks -> k-resident
+    # layout, ZERO broadcast panels; square configs split k
+    # only when m,k,n are comparable and ranks exceed the m/n splits.
+    assert E.summa_kresident(50000, 4096, 50000, 8)
+    assert E.summa_kresident(50000, 4096, 50000, 4)
+    assert E.summa_kresident(50000, 4096, 50000, 2)
+    # square 20000^3 on 8: splitMethod -> (2,2,2) -> k IS split -> slabbed
+    assert not E.summa_kresident(20000, 20000, 20000, 8)
+    # 1 rank: no distribution
+    assert not E.summa_kresident(50000, 4096, 50000, 1)
+    # mirrors the python restatement exactly for a sweep of shapes
+    from marlin_amd.api import split_method
+    for (m, k, n, c) in [(50000, 4096, 50000, 8), (100, 100, 100, 8),
+                         (7, 9000, 7, 8), (4096, 50000, 4096, 4),
+                         (20000, 20000, 20000, 2), (1, 50, 50, 8)]:
+        ms, ks, ns = split_method(m, k, n, c)
+        assert E.summa_kresident(m, k, n, c) == (c > 1 and ks == 1), \
+            (m, k, n, c)
+
+
+def test_kresident_dataflow_zero_exchange_virtual_ranks():
+    # Virtual-rank proof that the k-resident layout needs NO inter-rank
+    # exchange: every rank computes its C shard from ONLY its resident
+    # shards (row slab x full k, full k x col slab) and the assembly
+    # equals A @ B — config-4 shape (scaled), all grids incl. 8=4x2.
+    from oracle import gen_matrix
+    m, k, n = 500, 41, 460            # splitMethod never splits k=41
+    A = gen_matrix(m, k, seed=0xA11CE)
+    B = gen_matrix(k, n, seed=0xB0B)
+    ref = A @ B
+    for nranks in (2, 4, 8):
+        assert E.summa_kresident(m, k, n, nranks)
+        pr, pc = E.grid_shape(nranks)
+        C = np.full((m, n), np.nan)
+        for rank in range(nranks):
+            prow, pcol = rank // pc, rank % pc
+            mi, mo = E.slab_len(m, pr, prow), E.slab_off(m, pr, prow)
+            nj, no = E.slab_len(n, pc, pcol), E.slab_off(n, pc, pcol)
+            if mi == 0 or nj == 0:
+                continue
+            A_loc = A[mo:mo + mi, :]       # resident: ALL k columns
+            B_loc = B[:, no:no + nj]       # resident: ALL k rows
+            C[mo:mo + mi, no:no + nj] = A_loc @ B_loc
+        assert not np.isnan(C).any()
+        rel = np.max(np.abs(C - ref)) / np.max(np.abs(ref))
+        assert rel < 1e-13, (nranks, rel)
+
+
 def test_grid_shapes():
     assert E.grid_shape(8) == (4, 2)
     assert E.grid_shape(4) == (2, 2)
     assert E.grid_shape(2) == (2, 1)
     assert E.grid_shape(1) == (1, 1)
+
+
+def test_jni_veneer_compiles():
+    # src/host/marlin_jni.c must stay compilable for the day a JVM
+    # exists. No JDK here -> syntax-check against a
+    # minimal jni.h stub (test fixture, never shipped).
+    import subprocess
+    root = os.path.dirname(HERE)
+    src = os.path.join(root, "src", "host", "marlin_jni.c")
+    stub = os.path.join(HERE, "data", "jni_stub")
+    r = subprocess.run(
+        ["gcc", "-fsyntax-only", "-Wall", "-Werror", src,
+         "-I", os.path.join(root, "include"), "-I", stub],
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+
+
+def test_loaded_so_matches_build_hash():
+    # build() records the sha256 of the .so it just compiled; the library
+    # the tests load must be that exact binary (a stale committed binary
+    # must never mask a source regression)
+    rec = os.path.join(os.path.dirname(HERE), "marlin_amd", ".so_sha256")
+    if not os.path.exists(rec):
+        pytest.skip("no build hash recorded (build() not run)")
+    import hashlib
+    so = os.path.join(os.path.dirname(HERE), "marlin_amd",
+                      "libmarlin_gpu.so")
+    h = hashlib.sha256(open(so, "rb").read()).hexdigest()
+    assert h == open(rec).read().strip(), \
+        "libmarlin_gpu.so is not the binary build() produced"
 
 
 def test_strerror():

@@ -93,21 +93,81 @@ def _summa_rank(rank, world, m, k, n, q, port):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("world,mkn", [
-    (2, (96, 130, 64)), (2, (257, 99, 121)), (4, (150, 200, 120)),
-])
-def test_summa_dataflow_gloo(world, mkn):
-    m, k, n = mkn
+def _free_port():
+    # ephemeral port from the kernel: immune to TIME_WAIT collisions
+    # between parametrised cases / parallel runs
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _run_ranks(target, world, args):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29671 + (m % 100) + world
-    procs = [ctx.Process(target=_summa_rank,
-                         args=(r, world, m, k, n, q, port))
+    port = _free_port()
+    procs = [ctx.Process(target=target, args=(r, world) + args + (q, port))
              for r in range(world)]
     for p in procs:
         p.start()
     for p in procs:
         p.join(timeout=120)
         assert p.exitcode == 0
-    status, rel = q.get(timeout=10)
+    return q.get(timeout=10)
+
+
+@pytest.mark.parametrize("world,mkn", [
+    (2, (96, 130, 64)), (2, (257, 99, 121)), (4, (150, 200, 120)),
+])
+def test_summa_dataflow_gloo(world, mkn):
+    m, k, n = mkn
+    status, rel = _run_ranks(_summa_rank, world, mkn)
     assert status == "ok", f"SUMMA mismatch rel={rel}"
+
+
+def _kres_rank(rank, world, m, k, n, q, port):
+    # k-resident dataflow (config 4): each rank multiplies
+    # ONLY its resident shards — no broadcast is even issued; the
+    # assembled result must equal A @ B. Gloo is used solely to assemble.
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        assert E.summa_kresident(m, k, n, world)
+        pr, pc = E.grid_shape(world)
+        prow, pcol = rank // pc, rank % pc
+        A = gen_matrix(m, k, seed=0xA11CE)
+        B = gen_matrix(k, n, seed=0xB0B)
+        mi, mo = E.slab_len(m, pr, prow), E.slab_off(m, pr, prow)
+        nj, no = E.slab_len(n, pc, pcol), E.slab_off(n, pc, pcol)
+        A_local = A[mo:mo + mi, :]          # resident: ALL k columns
+        B_local = B[:, no:no + nj]          # resident: ALL k rows
+        C_local = A_local @ B_local          # zero inter-rank exchange
+        if rank == 0:
+            C = np.zeros((m, n))
+            C[mo:mo + mi, no:no + nj] = C_local
+            for r in range(1, world):
+                ri, rj = r // pc, r % pc
+                shp = (E.slab_len(m, pr, ri), E.slab_len(n, pc, rj))
+                t = torch.zeros(shp, dtype=torch.float64)
+                dist.recv(t, src=r)
+                ro, co = E.slab_off(m, pr, ri), E.slab_off(n, pc, rj)
+                C[ro:ro + shp[0], co:co + shp[1]] = t.numpy()
+            ref = A @ B
+            rel = np.max(np.abs(C - ref)) / np.max(np.abs(ref))
+            q.put(("ok", rel) if rel < 1e-13 else ("fail", rel))
+        else:
+            dist.send(torch.from_numpy(C_local.copy()), dst=0)
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("world,mkn", [
+    # config-4 shape scaled (k never split by splitMethod); ragged slabs
+    (2, (500, 41, 460)), (4, (500, 41, 460)), (4, (501, 96, 463)),
+])
+def test_kresident_dataflow_gloo(world, mkn):
+    m, k, n = mkn
+    status, rel = _run_ranks(_kres_rank, world, mkn)
+    assert status == "ok", f"k-resident mismatch rel={rel}"

@@ -507,6 +507,20 @@ class Engine:
             "mx_dgemv_device")
         return y[:dm.m].copy()
 
+    def dgemv_device_raw(self, m, n, dbuf, lda, x):
+        """y = A x on a raw device buffer (fp64, pitch lda): the
+        whole-matrix probe primitive for full-size parity tests."""
+        x = np.ascontiguousarray(x, dtype=np.float64)
+        assert x.shape[0] == n
+        y = np.empty(m, dtype=np.float64)
+        _ck(lib().mx_dgemv_device(self._ctx, m, n, dbuf, lda,
+                                  x.ctypes.data_as(ctypes.POINTER(
+                                      ctypes.c_double)),
+                                  y.ctypes.data_as(ctypes.POINTER(
+                                      ctypes.c_double))),
+            "mx_dgemv_device")
+        return y
+
     def transpose_dd(self, dm):
         """Device-resident transpose: returns a GEMM-ready DeviceMatrix
         (the padded image transposes wholesale; zero pads stay zero)."""

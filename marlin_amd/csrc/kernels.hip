@@ -77,7 +77,8 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
                                  const T* __restrict__ B, int64_t ldb,
                                  T* __restrict__ C, int64_t ldc,
                                  int nbm /* grid rows = M/BM */,
-                                 int band /* column-band width in blocks */) {
+                                 int band /* column-band width in blocks */,
+                                 int phase_mask /* k-phase stagger mask */) {
     constexpr int BN = 128;
     constexpr int NWAVES = WM * WN;
     constexpr int E = 16 / sizeof(T);       // elems per 16B glds chunk
@@ -227,16 +228,24 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
     }
 
     const int ntiles = (int)(K / BK);
-    issue_tile(0, 0);
+    // k-phase stagger experiment (MARLIN_GEMM_PHASE): blocks matching
+    // the mask start their k-loop half-way round (accumulation order
+    // rotated per tile — still deterministic run-to-run), so the two
+    // blocks sharing a CU do not hit their DMA-wait/barrier windows in
+    // sync. Default off (phase_mask 0 = ascending k, reference order).
+    const int ph = (phase_mask && (id & phase_mask)) ? (ntiles >> 1) : 0;
+    issue_tile(ph, 0);
 
     // Two barriers per k-step with the next tile's DMA issued EARLY and
     // kept in flight across the first barrier via a counted vmcnt (a
     // single-barrier variant with vmcnt(0)+late issue measured ~1% slower
     // fp64 / 4% slower fp32 - less DMA lead time).
-    for (int kt = 0; kt < ntiles; kt++) {
-        const int buf = kt & 1;
-        if (kt + 1 < ntiles) {
-            issue_tile(kt + 1, buf ^ 1);
+    for (int s = 0; s < ntiles; s++) {
+        const int buf = s & 1;
+        if (s + 1 < ntiles) {
+            int nxt = s + 1 + ph;
+            if (nxt >= ntiles) nxt -= ntiles;
+            issue_tile(nxt, buf ^ 1);
             // our own current-tile DMAs landed; next tile's stay in flight
             asm volatile("s_waitcnt vmcnt(%0)" :: "n"(A_GLDS + B_GLDS) : "memory");
         } else {
@@ -308,7 +317,11 @@ __global__ void gemm_mfma_kernel(int64_t M, int64_t N, int64_t K,
 }
 
 // fp32 GEMM with fused transpose/add epilogue (BASELINE config 5):
-// C_out[n x m] = (A*B)^T (+ addC). Same main loop; only the store differs.
+// C_out[n x m] = (A*B)^T (+ addC). Same main loop as the sgemm config
+// (incl. the f32 XOR swizzles and the supertiled band remap); the
+// transposed store is staged through LDS in two 64-row half-tiles so
+// every global write is a coalesced float4 run (the naive per-element
+// store was 4-byte scattered and cost ~8% at 40000^2).
 __launch_bounds__(256, 2)
 __global__ void sgemm_mfma_tn_epilogue_kernel(
         int64_t M, int64_t N, int64_t K,
@@ -318,13 +331,22 @@ __global__ void sgemm_mfma_tn_epilogue_kernel(
         const float* __restrict__ addC,          // N x M or nullptr
         int nbm, int band) {
     constexpr int BM = 128, BN = 128, BK = 16;
-    int nbn = (int)(N / BN);
+    // supertiled band remap (same as gemm_mfma_kernel: the 8 blocks on
+    // one XCD share a B tile column and a contiguous A panel in L2)
     int id = blockIdx.x;
-    int per_band = nbm * band;
-    int b0 = id / per_band;
-    int w = id - b0 * per_band;
-    int bn = b0 * band + w / nbm, bm = w % nbm;
-    (void)nbn;
+    int bn, bm;
+    {
+        int nbn = (int)(N / BN);
+        int per_band = nbm * band;
+        int b0 = id / per_band;
+        int w = id - b0 * per_band;
+        int first_bn = b0 * band;
+        int bw = min(band, nbn - first_bn);
+        int sh = 8 * bw;
+        int t = w / sh, l = w - t * sh;
+        bm = t * 8 + l / bw;
+        bn = first_bn + l % bw;
+    }
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -332,7 +354,7 @@ __global__ void sgemm_mfma_tn_epilogue_kernel(
     const int wm = wid >> 1, wn = wid & 1;
     const int l15 = lane & 15, l4 = lane >> 4;
 
-    __shared__ float sAB[2 * BK * BM + 2 * BN * BK];
+    __shared__ __align__(16) float sAB[2 * BK * BM + 2 * BN * BK];
     float* As = sAB;
     float* Bs = sAB + 2 * BK * BM;
     const int64_t row0 = (int64_t)bm * BM;
@@ -344,13 +366,17 @@ __global__ void sgemm_mfma_tn_epilogue_kernel(
     constexpr int B_COLS_PER_GLDS = (64 * E) / BK;
     constexpr int B_GLDS = BN / B_COLS_PER_GLDS / 4;
 
+    // f32 XOR swizzles on the GLOBAL source address (same as the sgemm
+    // config of gemm_mfma_kernel: A chunk ^4 by k parity, B chunk
+    // ^(c>>1)&3 — banks are mod 32 for b32 reads)
     auto issue_tile = [&](int kt, int buf) {
         const int64_t kbase = (int64_t)kt * BK;
         #pragma unroll
         for (int i = 0; i < A_GLDS; i++) {
             int c = (wid * A_GLDS + i) * A_COLS_PER_GLDS;
             int lane_col = lane / (BM / E);
-            int lane_row = (lane % (BM / E)) * E;
+            int lane_row = ((lane % (BM / E)) ^ (((c + lane_col) & 1) << 2))
+                           * E;
             glds16(A + (kbase + c + lane_col) * lda + row0 + lane_row,
                    &As[buf * BK * BM + c * BM]);
         }
@@ -358,7 +384,7 @@ __global__ void sgemm_mfma_tn_epilogue_kernel(
         for (int i = 0; i < B_GLDS; i++) {
             int c = (wid * B_GLDS + i) * B_COLS_PER_GLDS;
             int lane_col = lane / (BK / E);
-            int lane_row = (lane % (BK / E)) * E;
+            int lane_row = ((lane % 4) ^ (((c + lane_col) >> 1) & 3)) * E;
             glds16(B + (col0 + c + lane_col) * ldb + kbase + lane_row,
                    &Bs[buf * BN * BK + c * BK]);
         }
@@ -388,11 +414,17 @@ __global__ void sgemm_mfma_tn_epilogue_kernel(
             const int kk = q * 4 + l4;
             float a[4], b[4];
             #pragma unroll
-            for (int mt = 0; mt < 4; mt++)
-                a[mt] = At[kk * BM + wm * 64 + mt * 16 + l15];
+            for (int mt = 0; mt < 4; mt++) {
+                int rr = wm * 64 + mt * 16 + l15;
+                a[mt] = At[kk * BM + (((rr >> 2) ^ ((kk & 1) << 2)) << 2)
+                           + (rr & 3)];
+            }
             #pragma unroll
-            for (int nt = 0; nt < 4; nt++)
-                b[nt] = Bt[(wn * 64 + nt * 16 + l15) * BK + kk];
+            for (int nt = 0; nt < 4; nt++) {
+                int cb = wn * 64 + nt * 16 + l15;
+                b[nt] = Bt[cb * BK + (((kk >> 2) ^ ((cb >> 1) & 3)) << 2)
+                           + (kk & 3)];
+            }
             #pragma unroll
             for (int mt = 0; mt < 4; mt++)
                 #pragma unroll
@@ -403,20 +435,46 @@ __global__ void sgemm_mfma_tn_epilogue_kernel(
         __builtin_amdgcn_s_barrier();
     }
 
-    // transposed store: result element (r, c) of A*B goes to C[c + r*ldc];
-    // f32 16x16x4 row map: element j at row 4*l4 + j. addC fused.
+    // Transposed store via LDS: result element (r, cc) of A*B goes to
+    // C[cc + r*ldc] — contiguous output runs are (fixed r, varying cc).
+    // Stage 64 result rows at a time into sAB (exactly 64x128 floats),
+    // laid out stage[r'*128 + cc], then write float4 runs cooperatively.
+    float* stage = sAB;                 // k-loop done; reuse the buffers
+    v4f* stage4 = (v4f*)sAB;
     #pragma unroll
-    for (int mt = 0; mt < 4; mt++)
-        #pragma unroll
-        for (int nt = 0; nt < 4; nt++) {
-            int64_t r = row0 + wm * 64 + mt * 16 + l4 * 4;
-            int64_t cc = col0 + wn * 64 + nt * 16 + l15;
+    for (int half = 0; half < 2; half++) {
+        __builtin_amdgcn_s_barrier();   // prior phase's reads complete
+        if (wm == half) {
+            // this wave pair owns result rows half*64 .. half*64+63
             #pragma unroll
-            for (int j = 0; j < 4; j++) {
-                int64_t off = cc + (r + j) * ldc;
-                C[off] = acc[mt][nt][j] + (addC ? addC[off] : 0.0f);
-            }
+            for (int mt = 0; mt < 4; mt++)
+                #pragma unroll
+                for (int nt = 0; nt < 4; nt++) {
+                    int rloc = mt * 16 + l4 * 4;      // 0..63
+                    int cc = wn * 64 + nt * 16 + l15;
+                    #pragma unroll
+                    for (int j = 0; j < 4; j++)
+                        stage[(rloc + j) * BN + cc] = acc[mt][nt][j];
+                }
         }
+        __builtin_amdgcn_s_barrier();
+        // 64 rows x 32 float4 = 2048 stores over 256 threads
+        #pragma unroll
+        for (int i = 0; i < 8; i++) {
+            int idx = tid + i * 256;
+            int rloc = idx >> 5;                      // 0..63
+            int q = idx & 31;                         // float4 index in row
+            int64_t r = row0 + half * 64 + rloc;
+            int64_t off = col0 + 4 * q + r * ldc;
+            v4f v = stage4[rloc * (BN / 4) + q];
+            if (addC) {
+                const v4f* ap = (const v4f*)(addC + off);
+                v4f av = *ap;
+                v.x += av.x; v.y += av.y; v.z += av.z; v.w += av.w;
+            }
+            *(v4f*)(C + off) = v;
+        }
+    }
 }
 
 // ---------------------------------------------------------------------------
@@ -630,11 +688,14 @@ int mxk_gemm(int is_fp32, int beta_one,
     static const char* cfg = getenv("MARLIN_GEMM_CFG");
     bool bk32 = (K % 32 == 0) && cfg && cfg[0] == 'b' && cfg[1] == 'k';
     bool bm256 = (M % 256 == 0) && !is_fp32 && cfg && cfg[0] == 'b' && cfg[1] == 'm';
+    // k-phase stagger mask (experiment; see kernel comment). 0 = off.
+    static const char* phenv = getenv("MARLIN_GEMM_PHASE");
+    int phase_mask = phenv ? atoi(phenv) : 0;
     #define LAUNCH(TY, BETA, BK, BMv, WMv, WNv)                            \
         hipLaunchKernelGGL((gemm_mfma_kernel<TY, BETA, BK, BMv, WMv, WNv>),\
             dim3((unsigned)((M / BMv) * nbn)), dim3(WMv * WNv * 64), 0,    \
             stream, M, N, K, (const TY*)A, lda, (const TY*)B, ldb, (TY*)C, \
-            ldc, (int)(M / BMv), band)
+            ldc, (int)(M / BMv), band, phase_mask)
     if (is_fp32) {
         if (beta_one) { if (bk32) LAUNCH(float, 1, 32, 128, 2, 4); else LAUNCH(float, 1, 16, 128, 2, 2); }
         else          { if (bk32) LAUNCH(float, 0, 32, 128, 2, 4); else LAUNCH(float, 0, 16, 128, 2, 2); }

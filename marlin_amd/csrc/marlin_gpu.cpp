@@ -1041,8 +1041,10 @@ int mx_test_rccl_error(mx_ctx* c) {
   if (!c) return MX_EINVAL;
   if (!c->have_comm) return MX_ENOCOMM;
   HIP_OK(hipSetDevice(c->device));
-  ncclResult_t e = ncclBroadcast(nullptr, nullptr, 16, ncclFloat64, 0,
-                                 c->world, c->s_comm);
+  // out-of-range root: validated for ANY comm size (a null-buffer
+  // broadcast is short-circuited on a size-1 comm and never checked)
+  ncclResult_t e = ncclBroadcast(nullptr, nullptr, 16, ncclFloat64,
+                                 c->nranks + 7, c->world, c->s_comm);
   if (e != ncclSuccess) return MX_ERCCL;  // expected: invalid argument
   HIP_OK(hipStreamSynchronize(c->s_comm));
   return MX_OK;

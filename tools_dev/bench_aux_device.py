@@ -8,10 +8,11 @@ Each kernel runs PASSES times on a 16384^2 fp64 image (2.1 GB, 8x L3).
 Prints achieved GB/s from wall time (device-resident, no PCIe in the
 loop) against the 6.3 TB/s achievable HBM roofline.
 """
+import os
 import sys
 import time
 
-sys.path.insert(0, ".")
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import numpy as np  # noqa: E402
 

@@ -791,7 +791,10 @@ int mxk_transpose(int is_fp32, int64_t m, int64_t n, const void* in,
 int mxk_gemv(int is_fp32, int64_t m, int64_t n, int64_t lda, const void* A,
              const void* x, double* partial, void* y, hipStream_t stream) {
     int64_t nrb = (m + 255) / 256;
-    int nchunks = 8;
+    // 32 column chunks: at m=16384 that is 2048 workgroups (8 per CU),
+    // enough in-flight loads to cover HBM latency (8 chunks measured
+    // 0.53 TB/s -- only 2 blocks/CU; see profiles/r02_experiments.md)
+    int nchunks = 32;
     dim3 grid((unsigned)(nrb * nchunks)), block(256);
     if (is_fp32) {
         hipLaunchKernelGGL(gemv_partial_kernel<float>, grid, block, 0, stream,

@@ -966,7 +966,7 @@ int mx_dgemv(mx_ctx* c, int64_t m, int64_t n, const double* A,
   int rc;
   if ((rc = ensure(c, &c->wsA, m * n * 8))) return rc;
   if ((rc = ensure(c, &c->wsB, (n + m) * 8))) return rc;
-  if ((rc = ensure(c, &c->wsC, 8 * m * 8))) return rc;  // 8 chunk partials
+  if ((rc = ensure(c, &c->wsC, 32 * m * 8))) return rc;  // 32 chunk partials
   HIP_OK(hipMemcpyAsync(c->wsA.ptr, A, m * n * 8, hipMemcpyHostToDevice,
                         c->s_gemm));
   HIP_OK(hipMemcpyAsync(c->wsB.ptr, x, n * 8, hipMemcpyHostToDevice,
@@ -987,7 +987,7 @@ int mx_dgemv_device(mx_ctx* c, int64_t m, int64_t n, const mx_dbuf* dA,
   if (!c || !dA || !x || !y || m <= 0 || n <= 0) return MX_EINVAL;
   HIP_OK(hipSetDevice(c->device));
   int rc;
-  if ((rc = ensure(c, &c->wsB, (n + m + 8 * m) * 8))) return rc;
+  if ((rc = ensure(c, &c->wsB, (n + m + 32 * m) * 8))) return rc;
   double* dx = (double*)c->wsB.ptr;
   double* dy = dx + n;
   double* parts = dy + m;

@@ -30,6 +30,10 @@ timeout 400 $B --steps 3 > gpurun_out/r2c4_f64_base.log 2>&1
 MARLIN_GEMM_PHASE=8 timeout 400 $B --steps 3 > gpurun_out/r2c4_f64_ph8.log 2>&1
 MARLIN_GEMM_PHASE=1 timeout 400 $B --steps 3 > gpurun_out/r2c4_f64_ph1.log 2>&1
 
+# --- aux kernel bandwidth after gemv 32-chunk fix ---------------------
+timeout 300 python tools_dev/bench_aux_device.py \
+  > gpurun_out/r2c4_aux.log 2>&1
+
 # --- 2 ranks on 1 GPU (RCCL multi-rank smoke; bounded) ----------------
 export HSA_ENABLE_IPC_MODE_LEGACY=0
 MARLIN_FORCE_DEV0=1 MARLIN_BENCH_N=4096 MARLIN_SUMMA_DEBUG=1 \

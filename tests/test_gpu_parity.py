@@ -288,6 +288,40 @@ def test_full_size_random_columns_40000_fp32(eng):
         eng.free(dC)
 
 
+def test_probe_detects_single_element_corruption(eng):
+    # Sensitivity proof for the full-size random-vector probe: corrupt
+    # ONE element of an otherwise-correct product by 1 part in 1e8 and
+    # the probe must flag it (so a band-remap bug at any bn would be
+    # provably detectable by the 20000^3/40000^2 checks above).
+    m = k = n = 1024
+    a = gen_matrix(m, k, seed=0xDEAD)
+    b = gen_matrix(k, n, seed=0xBEEF)
+    A, B = eng.upload_matrix(a), eng.upload_matrix(b)
+    C = eng.gemm_dd(A, B)
+    ncap = (n + 127) // 128 * 128
+    kcap = (k + 127) // 128 * 128
+    v = np.zeros(ncap)
+    v[:n] = np.random.RandomState(3).rand(n)
+
+    def probe(cdev):
+        t = eng.dgemv_device_raw(B.pitch, ncap, B.buf, B.pitch, v)
+        ya = eng.dgemv_device_raw(A.pitch, kcap, A.buf, A.pitch,
+                                  np.pad(t, (0, kcap - B.pitch))
+                                  if kcap > B.pitch else t[:kcap])
+        yc = eng.dgemv_device_raw(cdev.pitch, ncap, cdev.buf, cdev.pitch, v)
+        return np.max(np.abs(yc - ya)) / np.max(np.abs(ya))
+
+    base = probe(C)
+    assert base < 1e-12, base
+    c_host = eng.download_matrix(C)
+    c_host[517, 709] *= 1 + 1e-8         # single-element fault
+    C2 = eng.upload_matrix(c_host)
+    corrupted = probe(C2)
+    assert corrupted > 100 * max(base, 1e-15), (base, corrupted)
+    for d in (A, B, C, C2):
+        d.free()
+
+
 @pytest.mark.parametrize("mn", [(100, 100), (517, 301), (4096, 1000), (1, 7)])
 def test_dgemv_parity(eng, mn):
     # BlockMatrix.multiply(DistributedVector/BDV) replacement (mx_dgemv)

@@ -17,7 +17,8 @@ def roundup(x, a):
 
 
 @pytest.mark.parametrize("nranks", [1, 2, 4, 8])
-@pytest.mark.parametrize("mkn", [(300, 500, 260), (129, 4097, 65)])
+@pytest.mark.parametrize("mkn", [(300, 500, 260), (129, 4097, 65),
+                                 (5, 300, 70)])  # m < pr: empty slabs
 def test_summa_virtual_ranks(nranks, mkn):
     m, k, n = mkn
     pr, pc = E.grid_shape(nranks)

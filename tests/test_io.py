@@ -68,3 +68,33 @@ def test_block_save_via_api_method(tmp_path):
     p2 = tmp_path / "dv.txt"
     blk.saveToFileSystem(str(p2))  # DenseVec format route
     np.testing.assert_array_equal(load_matrix_file(str(p2)).toBreeze(), a)
+
+
+def test_load_reference_generated_file(tmp_path):
+    """Parity anchor against the reference's OWN native generator:
+    oracle/_ref/generateMatrix is tools/generateMatrix.cpp compiled
+    unmodified from /root/reference (recipe: oracle/Makefile). Our
+    loader (MTUtils.loadMatrixFile semantics) must parse its output
+    exactly; values are U[0,5) floats (generateMatrix.cpp:14-24)."""
+    import subprocess
+    import pytest
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    ref_bin = os.path.join(root, "oracle", "_ref", "generateMatrix")
+    if not os.path.exists(ref_bin):
+        if not os.path.exists("/root/reference/tools/generateMatrix.cpp"):
+            pytest.skip("no reference tree and no prebuilt _ref binary")
+        subprocess.run(["make", "-C", os.path.join(root, "oracle"), "-s"],
+                       check=True)
+    out = subprocess.run([ref_bin, "37", "11"], capture_output=True,
+                         text=True, check=True).stdout
+    p = tmp_path / "ref_gen.txt"
+    p.write_text(out)
+    m = load_matrix_file(str(p))
+    a = m.toBreeze()
+    assert a.shape == (37, 11)
+    assert ((a >= 0) & (a < 5)).all()
+    # cross-parse with an independent minimal parser: exact float match
+    for line in out.strip().splitlines():
+        idx, data = line.split(":", 1)
+        vals = [float(v) for v in data.split(",")]
+        assert (a[int(idx)] == vals).all()

@@ -45,6 +45,9 @@ if [ "$MAX" -ge 8 ]; then
   MARLIN_GRID=2x4 run 8 n8_grid2x4
   MARLIN_SUMMA_KB=2048 run 8 n8_kb2048
   MARLIN_SUMMA_KB=8192 run 8 n8_kb8192
+  # bulk mode: panels = whole owner slabs (the all-gather-equivalent
+  # fallback SURVEY §8e names — all comm up front, no pipelining)
+  MARLIN_SUMMA_KB=20000 run 8 n8_kb_bulk
   # config 4: k-resident layout (zero steady-state xGMI traffic)
   MARLIN_BENCH_M=50000 MARLIN_BENCH_K=4096 MARLIN_BENCH_NN=50000 \
     run 8 n8_config4_kres

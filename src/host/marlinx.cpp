@@ -4,6 +4,9 @@
 //
 //   marlinx bench  <m> <k> <n> [steps=3] [warmup=1]   random fp64 multiply
 //   marlinx verify <m> <k> <n>                        engine vs CPU check
+//   marlinx epilogue <m> <k> <n>                      fp32 (A*B)^T + D check
+//     (config 5's fused transpose/add — BlockMatrix.scala:514-523 +
+//      :344-452 composed — through mx_sgemm_epilogue)
 //
 // The CPU check in `verify` is a naive triple loop on small sizes only —
 // a smoke-level verifier for the CLI; authoritative parity runs in
@@ -53,6 +56,37 @@ int main(int argc, char** argv) {
   if (rc != MX_OK) {
     fprintf(stderr, "mx_init: %s\n", mx_strerror(rc));
     return 1;
+  }
+
+  if (!strcmp(mode, "epilogue")) {
+    // fp32 fused (A*B)^T + D against a naive fp64 CPU recompute
+    std::vector<float> Af((size_t)(m * k)), Bf((size_t)(k * n)),
+        Cf((size_t)(n * m)), Df((size_t)(n * m));
+    for (int64_t i = 0; i < m * k; i++) Af[i] = (float)gen(0xA11CE, i);
+    for (int64_t i = 0; i < k * n; i++) Bf[i] = (float)gen(0xB0B, i);
+    for (int64_t i = 0; i < n * m; i++) Df[i] = (float)gen(0xADD, i);
+    rc = mx_sgemm_epilogue(ctx, m, k, n, Af.data(), Bf.data(), Cf.data(),
+                           /*transpose_c=*/1, Df.data());
+    if (rc != MX_OK) {
+      fprintf(stderr, "mx_sgemm_epilogue: %s\n", mx_strerror(rc));
+      return 1;
+    }
+    double maxrel = 0;
+    for (int64_t i = 0; i < m; i++)
+      for (int64_t j = 0; j < n; j++) {
+        double acc = 0;
+        for (int64_t l = 0; l < k; l++)
+          acc += (double)Af[l * m + i] * (double)Bf[j * k + l];
+        acc += (double)Df[i * n + j];            // D is n x m col-major
+        double d = (double)Cf[i * n + j] - acc;  // C_out is n x m
+        double rel = (d < 0 ? -d : d) / (acc < 0 ? -acc : acc);
+        if (rel > maxrel) maxrel = rel;
+      }
+    printf("epilogue %lldx%lldx%lld max_rel=%.3e %s\n", (long long)m,
+           (long long)k, (long long)n, maxrel,
+           maxrel < 1e-4 ? "OK" : "FAIL");
+    mx_shutdown(ctx);
+    return maxrel < 1e-4 ? 0 : 1;
   }
 
   if (!strcmp(mode, "verify")) {

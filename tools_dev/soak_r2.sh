@@ -13,7 +13,7 @@ echo "suite rc=$?" >> gpurun_out/${T}_suite.log
 python -c 'import __graft_entry__; __graft_entry__.smoke()' > gpurun_out/${T}_smoke.log 2>&1
 timeout 400 python bench.py --steps 3 --warmup 1 --no-cpu-baseline > gpurun_out/${T}_bench.log 2>&1
 timeout 400 python bench.py --workload epilogue --steps 2 --warmup 1 --no-cpu-baseline > gpurun_out/${T}_epi.log 2>&1
-timeout 600 python tools_dev/parity_sweep.py 100 > gpurun_out/${T}_sweep.log 2>&1
+timeout 600 python tools_dev/parity_sweep.py 100 $RANDOM > gpurun_out/${T}_sweep.log 2>&1
 ./marlinx verify 300 200 100 > gpurun_out/${T}_cli.log 2>&1
 ./marlinx epilogue 200 160 180 >> gpurun_out/${T}_cli.log 2>&1
 tail -2 gpurun_out/${T}_suite.log

@@ -88,6 +88,12 @@ def main():
                    default=env.get("MARLIN_BENCH_WORKLOAD", "gemm"))
     p.add_argument("--cpu-sample", type=int, default=12000)
     p.add_argument("--no-cpu-baseline", action="store_true")
+    # run the SUMMA panel-loop path even at world 1 (1x1 grid, RCCL comm
+    # of size 1): measures the distributed pipeline's overhead (panel
+    # packs + beta-chained GEMMs) without xGMI traffic — the pre-SCALE
+    # machinery check
+    p.add_argument("--force-summa", action="store_true",
+                   default=bool(env.get("MARLIN_BENCH_FORCE_SUMMA")))
     args = p.parse_args()
 
     if args.workload == "epilogue":
@@ -169,6 +175,21 @@ def main():
 
             def step():
                 summa(m, k, n, dA, dB, dC)
+    elif args.force_summa:
+        # 1x1-grid SUMMA: the full panel-loop machinery, zero comm
+        eng.comm_init(0, 1, Engine.comm_id())
+        kbi_p1 = roundup(k, 16)
+        dA = eng.alloc(mip * kaj * elem)       # mi=m, kaj=k at 1x1
+        dB = eng.alloc(kbi_p1 * nj * elem)
+        dC = eng.alloc(mip * njp * elem)
+        eng.fill_random(dA, mip * kaj, 0xA11CE, fp32)
+        eng.fill_random(dB, kbi_p1 * nj, 0xB0B, fp32)
+        eng.zero_pad(dA, mip, kaj, mip, mi, kaj, fp32)
+        eng.zero_pad(dB, kbi_p1, nj, kbi_p1, kbi, nj, fp32)
+        summa = eng.sgemm_summa_device if fp32 else eng.dgemm_summa_device
+
+        def step():
+            summa(m, k, n, dA, dB, dC)
     else:
         mp, kp, np_ = roundup(m, 128), roundup(k, 16), roundup(n, 128)
         dA = eng.alloc(mp * kp * elem)
@@ -274,6 +295,8 @@ def main():
         if world > 1:
             par = (f"kres_grid_{pr}x{pc}" if kres else
                    f"summa_grid_{pr}x{pc}")
+        elif args.force_summa:
+            par = "summa_1x1_machinery"
         else:
             par = "single_gpu"
         out = {

@@ -625,10 +625,13 @@ static int summa_device(mx_ctx* c, int is_fp32, int64_t m, int64_t k,
   const bool has_tile = mip > 0 && njp > 0;
   // local shard pitches ARE the padded sizes (bench fills them that way;
   // host entry packs them that way)
-  // panel width: overlap granularity of the comm/MFMA pipeline
-  // (MARLIN_SUMMA_KB for round-2 8-GPU tuning; default 4096)
+  // panel width: overlap granularity of the comm/MFMA pipeline.
+  // Default 2048 (measured: the 1-rank panel-loop machinery runs at
+  // 98.9% of the monolithic GEMM rate at KB=2048 vs 97.2% at 4096 —
+  // profiles/r02_experiments.md; broadcasts still hide under the
+  // ~3 ms per-panel GEMM at 8 GPUs). MARLIN_SUMMA_KB overrides.
   static const char* kbenv = getenv("MARLIN_SUMMA_KB");
-  const int64_t kb_max = kbenv && atoll(kbenv) > 0 ? atoll(kbenv) : 4096;
+  const int64_t kb_max = kbenv && atoll(kbenv) > 0 ? atoll(kbenv) : 2048;
   auto panels = plan_panels(k, c->pr, c->pc, kb_max);
 
   c->st = {};
